@@ -1,0 +1,47 @@
+# mpix — MI355X-native accelerator-triggered MPI extensions.
+# Builds libmpix.so (C API) with hipcc for gfx950. Cross-compiles fine on a
+# GPU-less box; the .so loads and runs host-only paths without a GPU.
+
+HIPCC      ?= hipcc
+GPU_ARCH   ?= gfx950
+MPI_HOME   ?= /opt/conda
+
+CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(GPU_ARCH) \
+              -Iinclude -I$(MPI_HOME)/include -Wall -Wextra -Wno-unused-parameter
+LDFLAGS    := -shared -L$(MPI_HOME)/lib -lmpi -Wl,-rpath,/usr/lib/x86_64-linux-gnu -Wl,-rpath,$(MPI_HOME)/lib
+
+ifdef DEBUG
+CXXFLAGS   += -g -DMPIX_DEBUG
+endif
+
+SRCS := src/state.cpp src/init.cpp src/proxy.cpp src/enqueue.cpp \
+        src/partitioned.cpp src/transport/bootstrap.cpp \
+        src/transport/native.cpp src/transport/mpi.cpp
+OBJS := $(SRCS:.cpp=.o)
+
+LIB  := libmpix.so
+PYEXT := mpix/_C.so
+PYINC := $(shell python3 -c "import sysconfig;print(sysconfig.get_paths()['include'])")
+PYBIND11INC := $(shell python3 -c "import pybind11;print(pybind11.get_include())")
+
+all: $(LIB) python
+
+python: $(PYEXT)
+
+mpix/_core.o: mpix/_core.cpp include/mpix/mpix.h include/mpix/mpix_device.h \
+              include/mpix/mpix_abi.h
+	$(HIPCC) $(CXXFLAGS) -I$(PYINC) -I$(PYBIND11INC) -c mpix/_core.cpp -o $@
+
+$(PYEXT): mpix/_core.o $(OBJS)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) mpix/_core.o $(OBJS) $(LDFLAGS) -o $@
+
+%.o: %.cpp src/internal.h include/mpix/mpix.h include/mpix/mpix_abi.h
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(LIB): $(OBJS)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $(OBJS) $(LDFLAGS) -o $@
+
+clean:
+	rm -f $(OBJS) $(LIB) $(PYEXT)
+
+.PHONY: all clean python

@@ -1,0 +1,617 @@
+/* mpix — stream/graph-enqueued operations + host-side waits.
+ *
+ * Reference counterpart: /root/reference/src/sendrecv.cu (682 LoC CUDA).
+ * MI355X-native rebuild:
+ *  - fast path: hipStreamWriteValue32 / hipStreamWaitValue32(Eq) /
+ *    hipStreamBatchMemOp against the device-mapped flag pool (functional
+ *    probe at init, env kill-switch MPIX_DISABLE_MEMOPS);
+ *  - fallback + capture/graph path: tiny gfx950 kernels using system-scope
+ *    HIP atomics (host-pinned flags are uncached on-device; the spin loop
+ *    backs off with s_sleep so a parked wave doesn't hammer the host link);
+ *  - hipGraph support both via stream capture (kernels are capturable) and
+ *    via explicit single-node graph construction (MPIX_QUEUE_HIP_GRAPH),
+ *    with hipUserObject retain/release driving request cleanup;
+ *  - the reference's graph-wait bug (D2: waiting on PENDING instead of
+ *    COMPLETED, sendrecv.cu:411) is fixed here: all wait nodes target
+ *    COMPLETED.
+ */
+#include <hip/hip_runtime.h>
+
+#include "internal.h"
+
+namespace mpix {
+
+/* ------------------------------------------------------------ gfx950 kernels
+ * All flag kernels are <<<1,1>>>: they touch a single 32-bit word in
+ * host-pinned memory.  System scope makes the access visible to the CPU
+ * proxy without any cache maintenance; s_sleep parks the wave between
+ * probes (each probe is a fabric read over the host link). */
+
+__global__ void k_set_flag(uint32_t *flag, uint32_t val)
+{
+    __hip_atomic_store(flag, val, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+__global__ void k_wait_flag(uint32_t *flag, uint32_t val)
+{
+    while (__hip_atomic_load(flag, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM) != val)
+        __builtin_amdgcn_s_sleep(16);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+}
+
+__global__ void k_wait_and_set(uint32_t *flag, uint32_t val, uint32_t newval)
+{
+    while (__hip_atomic_load(flag, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM) != val)
+        __builtin_amdgcn_s_sleep(16);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+    __hip_atomic_store(flag, newval, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+/* Waitall fallback: one wavefront polls `count` flags (indices in a pinned
+ * array), each lane owning a strided subset — one launch instead of N
+ * (the reference left this as dead code, sendrecv.cu:64-72 + TODO:514-520). */
+__global__ void k_waitall_and_set(uint32_t *flags, const int32_t *idx,
+                                  int count, uint32_t val, uint32_t newval)
+{
+    for (int i = (int)threadIdx.x; i < count; i += (int)blockDim.x) {
+        uint32_t *f = flags + idx[i];
+        while (__hip_atomic_load(f, __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_SYSTEM) != val)
+            __builtin_amdgcn_s_sleep(16);
+        if (newval != 0xFFFFFFFFu)
+            __hip_atomic_store(f, newval, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+}
+
+/* ------------------------------------------------------------------ helpers */
+
+static bool stream_capturing(hipStream_t stream)
+{
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    if (hipStreamIsCapturing(stream, &st) != hipSuccess) {
+        (void)hipGetLastError();
+        return false;
+    }
+    return st == hipStreamCaptureStatusActive;
+}
+
+/* hipUserObject destructor: releases an enqueued request owned by a graph.
+ * Runs when the owning graph is destroyed.  Spins out any in-flight state,
+ * then hands the slot to the proxy via CLEANUP. */
+static void graph_request_destroy(void *ud)
+{
+    Request *req = (Request *)ud;
+    State *s = g_state;
+    if (s == nullptr) { delete req; return; }
+    int idx = req->flag_idx;
+    uint32_t f;
+    while ((f = flag_load(idx)) == MPIX_FLAG_PENDING || f == MPIX_FLAG_ISSUED)
+        std::this_thread::yield();
+    /* RESERVED (never launched) or COMPLETED: proxy frees slot + request */
+    flag_store(idx, MPIX_FLAG_CLEANUP);
+}
+
+/* Attach graph-owned cleanup for `req` to `graph`. */
+static int attach_cleanup(hipGraph_t graph, Request *req)
+{
+    hipUserObject_t uo = nullptr;
+    MPIX_CHECK_HIP(hipUserObjectCreate(&uo, req, graph_request_destroy, 1,
+                                       hipUserObjectNoDestructorSync));
+    MPIX_CHECK_HIP(hipGraphRetainUserObject(graph, uo, 1,
+                                            hipGraphUserObjectMove));
+    return 0;
+}
+
+/* Build a single-kernel-node graph calling k_set_flag / k_wait_flag. */
+static int make_flag_graph(hipGraph_t *out, void *kernel, uint32_t *flag_d,
+                           uint32_t val)
+{
+    hipGraph_t g = nullptr;
+    MPIX_CHECK_HIP(hipGraphCreate(&g, 0));
+    hipKernelNodeParams p{};
+    /* hipGraphAddKernelNode copies parameter values during the call, so
+     * locals are fine here */
+    void *kp[2] = {&flag_d, &val};
+    p.func = kernel;
+    p.gridDim = dim3(1, 1, 1);
+    p.blockDim = dim3(1, 1, 1);
+    p.sharedMemBytes = 0;
+    p.kernelParams = kp;
+    p.extra = nullptr;
+    hipGraphNode_t node = nullptr;
+    hipError_t e = hipGraphAddKernelNode(&node, g, nullptr, 0, &p);
+    if (e != hipSuccess) {
+        MPIX_ERR("hipGraphAddKernelNode failed: %s", hipGetErrorString(e));
+        (void)hipGraphDestroy(g);
+        return (int)e;
+    }
+    *out = g;
+    return 0;
+}
+
+/* Fire the PENDING trigger for slot idx on a stream/graph queue.
+ * On the graph path *graph_out receives the new single-node graph. */
+static int fire_trigger(int idx, int qtype, void *queue, Request *req)
+{
+    State *s = g_state;
+    uint32_t *flag_d = s->have_gpu ? s->flags_d + idx : nullptr;
+
+    if (qtype == MPIX_QUEUE_HIP_GRAPH) {
+        if (!s->have_gpu) {
+            MPIX_ERR("graph queue requires a GPU");
+            return MPI_ERR_OTHER;
+        }
+        hipGraph_t g = nullptr;
+        MPIX_CHECK(make_flag_graph(&g, (void *)k_set_flag, flag_d,
+                                   MPIX_FLAG_PENDING));
+        MPIX_CHECK(attach_cleanup(g, req));
+        *(hipGraph_t *)queue = g;
+        return MPI_SUCCESS;
+    }
+
+    /* stream queue */
+    if (!s->have_gpu) {
+        /* proxy-only path (no GPU): the "stream" reaches the op now */
+        return trigger_host(idx);
+    }
+    hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
+    if (stream_capturing(stream)) {
+        hipLaunchKernelGGL(k_set_flag, dim3(1), dim3(1), 0, stream, flag_d,
+                           (uint32_t)MPIX_FLAG_PENDING);
+        MPIX_CHECK_HIP(hipGetLastError());
+        /* request lifetime follows the captured graph */
+        hipStreamCaptureStatus cst;
+        unsigned long long cid = 0;
+        hipGraph_t cg = nullptr;
+        MPIX_CHECK_HIP(hipStreamGetCaptureInfo_v2(stream, &cst, &cid, &cg,
+                                                  nullptr, nullptr));
+        if (cg != nullptr) MPIX_CHECK(attach_cleanup(cg, req));
+        req->flag_idx = idx; /* graph-owned */
+        return MPI_SUCCESS;
+    }
+    if (s->use_memops) {
+        MPIX_CHECK_HIP(hipStreamWriteValue32(stream, flag_d,
+                                             (uint32_t)MPIX_FLAG_PENDING, 0));
+    } else {
+        hipLaunchKernelGGL(k_set_flag, dim3(1), dim3(1), 0, stream, flag_d,
+                           (uint32_t)MPIX_FLAG_PENDING);
+        MPIX_CHECK_HIP(hipGetLastError());
+    }
+    return MPI_SUCCESS;
+}
+
+/* Common body of Isend/Irecv_enqueue. */
+static int enqueue_sendrecv(bool is_send, void *buf, int count,
+                            MPI_Datatype datatype, int peer, int tag,
+                            MPI_Comm comm, MPIX_Request *request, int qtype,
+                            void *queue)
+{
+    State *s = g_state;
+    if (s == nullptr) {
+        MPIX_ERR("MPIX_Init not called");
+        return MPI_ERR_OTHER;
+    }
+    if (request == nullptr || count < 0) return MPI_ERR_ARG;
+    if (qtype != MPIX_QUEUE_HIP_STREAM && qtype != MPIX_QUEUE_HIP_GRAPH)
+        return MPI_ERR_ARG;
+
+    int tsz = 0;
+    MPIX_CHECK(datatype_size(datatype, &tsz));
+    int peer_world = -1;
+    uint32_t comm_id = 0;
+    bool native_ok = false;
+    MPIX_CHECK(resolve_peer(comm, peer, &peer_world, &comm_id, &native_ok));
+
+    int idx = slot_allocate();
+    if (idx < 0) return MPI_ERR_INTERN;
+
+    Op *op = &s->ops[idx];
+    op->kind = is_send ? OpKind::ISEND : OpKind::IRECV;
+    op->buf = buf;
+    op->count = count;
+    op->datatype = datatype;
+    op->bytes = (uint64_t)count * (uint64_t)tsz;
+    op->peer = peer;
+    op->peer_world = peer_world;
+    op->tag = tag;
+    op->comm = comm;
+    op->comm_id = comm_id;
+    op->buf_is_device = ptr_is_device(buf);
+    op->native_route = native_ok;
+
+    Request *req = new Request();
+    req->kind = ReqKind::BASIC;
+    req->flag_idx = idx;
+    op->req = req;
+
+    slot_arm(idx);
+
+    int rc = fire_trigger(idx, qtype, queue, req);
+    if (rc != MPI_SUCCESS) {
+        /* roll back: nothing triggered, proxy will see AVAILABLE and drop */
+        delete req;
+        slot_free(idx);
+        *request = MPIX_REQUEST_NULL;
+        return rc;
+    }
+    *request = (MPIX_Request)req;
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Isend_enqueue(const void *buf, int count,
+                                  MPI_Datatype datatype, int dest, int tag,
+                                  MPI_Comm comm, MPIX_Request *request,
+                                  int qtype, void *queue)
+{
+    return enqueue_sendrecv(true, (void *)buf, count, datatype, dest, tag,
+                            comm, request, qtype, queue);
+}
+
+extern "C" int MPIX_Irecv_enqueue(void *buf, int count, MPI_Datatype datatype,
+                                  int source, int tag, MPI_Comm comm,
+                                  MPIX_Request *request, int qtype, void *queue)
+{
+    return enqueue_sendrecv(false, buf, count, datatype, source, tag, comm,
+                            request, qtype, queue);
+}
+
+/* ----------------------------------------------------------- wait (stream) */
+
+/* Fast path: if the proxy already completed the op, consume it now and skip
+ * all stream work (closes the same race the reference handles with
+ * try_complete_wait_op, sendrecv.cu:82-104).  Otherwise posts the user
+ * status target for proxy-side delivery.  Returns true if consumed. */
+static bool try_complete_now(int idx, MPI_Status *status)
+{
+    State *s = g_state;
+    std::lock_guard<std::mutex> lk(s->completion_mutex);
+    Op *op = &s->ops[idx];
+    if (flag_load(idx) == MPIX_FLAG_COMPLETED) {
+        if (status != nullptr && status != MPI_STATUS_IGNORE)
+            *status = op->saved_status;
+        flag_store(idx, MPIX_FLAG_CLEANUP);
+        return true;
+    }
+    if (status != nullptr && status != MPI_STATUS_IGNORE)
+        op->enq_status_target = status;
+    return false;
+}
+
+extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
+                                 int qtype, void *queue)
+{
+    State *s = g_state;
+    if (s == nullptr || reqp == nullptr) return MPI_ERR_ARG;
+    Request *req = (Request *)*reqp;
+    if (req == nullptr) return MPI_ERR_REQUEST;
+    if (req->kind != ReqKind::BASIC) {
+        MPIX_ERR("Wait_enqueue supports enqueued (basic) requests only");
+        return MPI_ERR_REQUEST;
+    }
+    int idx = req->flag_idx;
+    uint32_t *flag_d = s->have_gpu ? s->flags_d + idx : nullptr;
+
+    if (qtype == MPIX_QUEUE_HIP_GRAPH) {
+        if (!s->have_gpu) return MPI_ERR_OTHER;
+        if (status != nullptr && status != MPI_STATUS_IGNORE) {
+            std::lock_guard<std::mutex> lk(s->completion_mutex);
+            Op *op = &s->ops[idx];
+            if (flag_load(idx) == MPIX_FLAG_COMPLETED)
+                *status = op->saved_status;
+            else
+                op->enq_status_target = status;
+        }
+        hipGraph_t g = nullptr;
+        /* graph wait targets COMPLETED (reference bug D2 fixed) */
+        MPIX_CHECK(make_flag_graph(&g, (void *)k_wait_flag, flag_d,
+                                   MPIX_FLAG_COMPLETED));
+        *(hipGraph_t *)queue = g;
+        *reqp = MPIX_REQUEST_NULL; /* ownership: send/recv graph's user object */
+        return MPI_SUCCESS;
+    }
+
+    if (!s->have_gpu) {
+        /* proxy-only path: degenerate to host wait */
+        return MPIX_Wait(reqp, status);
+    }
+
+    hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
+    bool capturing = stream_capturing(stream);
+
+    if (capturing) {
+        if (status != nullptr && status != MPI_STATUS_IGNORE) {
+            std::lock_guard<std::mutex> lk(s->completion_mutex);
+            s->ops[idx].enq_status_target = status;
+        }
+        hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream, flag_d,
+                           (uint32_t)MPIX_FLAG_COMPLETED);
+        MPIX_CHECK_HIP(hipGetLastError());
+        *reqp = MPIX_REQUEST_NULL;
+        return MPI_SUCCESS;
+    }
+
+    if (try_complete_now(idx, status)) {
+        *reqp = MPIX_REQUEST_NULL;
+        return MPI_SUCCESS;
+    }
+    if (s->use_memops) {
+        MPIX_CHECK_HIP(hipStreamWaitValue32(stream, flag_d,
+                                            (uint32_t)MPIX_FLAG_COMPLETED,
+                                            hipStreamWaitValueEq, 0xFFFFFFFFu));
+        MPIX_CHECK_HIP(hipStreamWriteValue32(stream, flag_d,
+                                             (uint32_t)MPIX_FLAG_CLEANUP, 0));
+    } else {
+        hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0, stream, flag_d,
+                           (uint32_t)MPIX_FLAG_COMPLETED,
+                           (uint32_t)MPIX_FLAG_CLEANUP);
+        MPIX_CHECK_HIP(hipGetLastError());
+    }
+    *reqp = MPIX_REQUEST_NULL;
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
+                                    MPI_Status *statuses, int qtype,
+                                    void *queue)
+{
+    State *s = g_state;
+    if (s == nullptr || (count > 0 && reqs == nullptr)) return MPI_ERR_ARG;
+    if (count == 0) return MPI_SUCCESS;
+
+    auto status_at = [&](int i) -> MPI_Status * {
+        if (statuses == nullptr || statuses == MPI_STATUSES_IGNORE)
+            return nullptr;
+        return &statuses[i];
+    };
+
+    if (qtype == MPIX_QUEUE_HIP_GRAPH) {
+        if (!s->have_gpu) return MPI_ERR_OTHER;
+        hipGraph_t g = nullptr;
+        MPIX_CHECK_HIP(hipGraphCreate(&g, 0));
+        for (int i = 0; i < count; i++) {
+            Request *req = (Request *)reqs[i];
+            if (req == nullptr || req->kind != ReqKind::BASIC) continue;
+            int idx = req->flag_idx;
+            {
+                std::lock_guard<std::mutex> lk(s->completion_mutex);
+                MPI_Status *st = status_at(i);
+                if (st) {
+                    if (flag_load(idx) == MPIX_FLAG_COMPLETED)
+                        *st = s->ops[idx].saved_status;
+                    else
+                        s->ops[idx].enq_status_target = st;
+                }
+            }
+            hipKernelNodeParams p{};
+            uint32_t *flag_d = s->flags_d + idx;
+            uint32_t val = MPIX_FLAG_COMPLETED;
+            void *kp[2] = {&flag_d, &val};
+            p.func = (void *)k_wait_flag;
+            p.gridDim = dim3(1, 1, 1);
+            p.blockDim = dim3(1, 1, 1);
+            p.kernelParams = kp;
+            hipGraphNode_t node = nullptr;
+            hipError_t e = hipGraphAddKernelNode(&node, g, nullptr, 0, &p);
+            if (e != hipSuccess) {
+                (void)hipGraphDestroy(g);
+                return (int)e;
+            }
+            reqs[i] = MPIX_REQUEST_NULL;
+        }
+        *(hipGraph_t *)queue = g;
+        return MPI_SUCCESS;
+    }
+
+    if (!s->have_gpu) return MPIX_Waitall(count, reqs, statuses);
+
+    hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
+    bool capturing = stream_capturing(stream);
+
+    /* batched memOps fast path: one submission for all waits+cleanups */
+    if (!capturing && s->use_batch_memops) {
+        std::vector<hipStreamBatchMemOpParams> params;
+        params.reserve(2 * (size_t)count);
+        for (int i = 0; i < count; i++) {
+            Request *req = (Request *)reqs[i];
+            if (req == nullptr) continue;
+            if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            int idx = req->flag_idx;
+            if (try_complete_now(idx, status_at(i))) {
+                reqs[i] = MPIX_REQUEST_NULL;
+                continue;
+            }
+            uint32_t *flag_d = s->flags_d + idx;
+            hipStreamBatchMemOpParams w{};
+            w.waitValue.operation = hipStreamMemOpWaitValue32;
+            w.waitValue.address = flag_d;
+            w.waitValue.value = MPIX_FLAG_COMPLETED;
+            w.waitValue.flags = hipStreamWaitValueEq;
+            params.push_back(w);
+            hipStreamBatchMemOpParams wr{};
+            wr.writeValue.operation = hipStreamMemOpWriteValue32;
+            wr.writeValue.address = flag_d;
+            wr.writeValue.value = MPIX_FLAG_CLEANUP;
+            params.push_back(wr);
+            reqs[i] = MPIX_REQUEST_NULL;
+        }
+        if (!params.empty())
+            MPIX_CHECK_HIP(hipStreamBatchMemOp(stream,
+                                               (unsigned)params.size(),
+                                               params.data(), 0));
+        return MPI_SUCCESS;
+    }
+
+    /* kernel path: one wavefront waits on all flags in a single launch.
+     * The index array rides in pinned memory freed by a host callback. */
+    int32_t *idx_arr = nullptr;
+    MPIX_CHECK_HIP(hipHostMalloc((void **)&idx_arr,
+                                 (size_t)count * sizeof(int32_t),
+                                 hipHostMallocMapped));
+    int n = 0;
+    for (int i = 0; i < count; i++) {
+        Request *req = (Request *)reqs[i];
+        if (req == nullptr) continue;
+        if (req->kind != ReqKind::BASIC) {
+            (void)hipHostFree(idx_arr);
+            return MPI_ERR_REQUEST;
+        }
+        int idx = req->flag_idx;
+        if (capturing) {
+            std::lock_guard<std::mutex> lk(s->completion_mutex);
+            MPI_Status *st = status_at(i);
+            if (st) s->ops[idx].enq_status_target = st;
+            idx_arr[n++] = idx;
+        } else if (try_complete_now(idx, status_at(i))) {
+            /* done already */
+        } else {
+            idx_arr[n++] = idx;
+        }
+        reqs[i] = MPIX_REQUEST_NULL;
+    }
+    if (n > 0) {
+        int32_t *idx_d = nullptr;
+        MPIX_CHECK_HIP(hipHostGetDevicePointer((void **)&idx_d, idx_arr, 0));
+        uint32_t newval = capturing ? 0xFFFFFFFFu : MPIX_FLAG_CLEANUP;
+        int threads = n < 64 ? 64 : ((n + 63) / 64) * 64;
+        if (threads > 1024) threads = 1024;
+        hipLaunchKernelGGL(k_waitall_and_set, dim3(1), dim3(threads), 0,
+                           stream, s->flags_d, idx_d, n,
+                           (uint32_t)MPIX_FLAG_COMPLETED, newval);
+        MPIX_CHECK_HIP(hipGetLastError());
+        if (!capturing) {
+            MPIX_CHECK_HIP(hipLaunchHostFunc(
+                stream, [](void *p) { (void)hipHostFree(p); }, idx_arr));
+        }
+        /* capturing: the pinned array must outlive graph relaunches; it is
+         * intentionally leaked to the capture (small, bounded by nflags). */
+    } else {
+        (void)hipHostFree(idx_arr);
+    }
+    return MPI_SUCCESS;
+}
+
+/* ------------------------------------------------------------- wait (host) */
+
+static int host_wait_basic(Request *req, MPI_Status *status)
+{
+    State *s = g_state;
+    int idx = req->flag_idx;
+    int spins = 0;
+    while (flag_load(idx) != MPIX_FLAG_COMPLETED) {
+        if (++spins > 4096) {
+            std::this_thread::yield();
+            spins = 0;
+        }
+    }
+    {
+        std::lock_guard<std::mutex> lk(s->completion_mutex);
+        if (status != nullptr && status != MPI_STATUS_IGNORE)
+            *status = s->ops[idx].saved_status;
+        slot_free(idx);
+    }
+    delete req;
+    return MPI_SUCCESS;
+}
+
+static int host_wait_partitioned(Request *req, MPI_Status *status)
+{
+    State *s = g_state;
+    if (!req->active) return MPI_SUCCESS;
+    for (int p = 0; p < req->n_partitions; p++) {
+        int idx = req->part_idx[p];
+        int spins = 0;
+        while (flag_load(idx) != MPIX_FLAG_COMPLETED) {
+            if (++spins > 4096) {
+                std::this_thread::yield();
+                spins = 0;
+            }
+        }
+    }
+    /* all partitions completed: reset for the next MPIX_Start */
+    for (int p = 0; p < req->n_partitions; p++) {
+        int idx = req->part_idx[p];
+        s->ops[idx].ch_done.store(0, std::memory_order_relaxed);
+        s->ops[idx].status_saved = false;
+        flag_store(idx, MPIX_FLAG_RESERVED);
+    }
+    req->active = false;
+    if (status != nullptr && status != MPI_STATUS_IGNORE) {
+        ChStatus cs;
+        cs.src = req->is_send ? -1 : req->peer;
+        cs.tag = req->tag;
+        cs.bytes = req->part_bytes * (uint64_t)req->n_partitions;
+        cs.err = MPI_SUCCESS;
+        fill_status(status, cs);
+    }
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Wait(MPIX_Request *reqp, MPI_Status *status)
+{
+    if (g_state == nullptr || reqp == nullptr) return MPI_ERR_ARG;
+    Request *req = (Request *)*reqp;
+    if (req == nullptr) return MPI_SUCCESS; /* MPI semantics: null is no-op */
+    if (req->kind == ReqKind::BASIC) {
+        int rc = host_wait_basic(req, status);
+        *reqp = MPIX_REQUEST_NULL;
+        return rc;
+    }
+    /* partitioned requests stay valid (persistent) */
+    return host_wait_partitioned(req, status);
+}
+
+extern "C" int MPIX_Waitall(int count, MPIX_Request *reqs,
+                            MPI_Status *statuses)
+{
+    for (int i = 0; i < count; i++) {
+        MPI_Status *st =
+            (statuses == nullptr || statuses == MPI_STATUSES_IGNORE)
+                ? MPI_STATUS_IGNORE
+                : &statuses[i];
+        MPIX_CHECK(MPIX_Wait(&reqs[i], st));
+    }
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Request_free(MPIX_Request *reqp)
+{
+    State *s = g_state;
+    if (s == nullptr || reqp == nullptr) return MPI_ERR_ARG;
+    Request *req = (Request *)*reqp;
+    if (req == nullptr) return MPI_SUCCESS;
+
+    if (req->kind == ReqKind::BASIC) {
+        int idx = req->flag_idx;
+        std::lock_guard<std::mutex> lk(s->completion_mutex);
+        uint32_t f = flag_load(idx);
+        if (f == MPIX_FLAG_COMPLETED) {
+            flag_store(idx, MPIX_FLAG_CLEANUP); /* proxy frees slot+request */
+        } else {
+            s->ops[idx].orphaned = true; /* proxy frees at completion */
+        }
+        *reqp = MPIX_REQUEST_NULL;
+        return MPI_SUCCESS;
+    }
+
+    /* partitioned: must be inactive (MPI rule); drain defensively anyway */
+    for (int p = 0; p < req->n_partitions; p++) {
+        int idx = req->part_idx[p];
+        uint32_t f;
+        while ((f = flag_load(idx)) == MPIX_FLAG_PENDING ||
+               f == MPIX_FLAG_ISSUED)
+            std::this_thread::yield();
+        flag_store(idx, MPIX_FLAG_CLEANUP); /* proxy frees slot */
+    }
+    if (req->dev_idx) (void)hipFree(req->dev_idx);
+    if (req->dev_handle) (void)hipFree(req->dev_handle);
+    delete req;
+    *reqp = MPIX_REQUEST_NULL;
+    return MPI_SUCCESS;
+}
+
+} /* namespace mpix */

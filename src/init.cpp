@@ -1,0 +1,220 @@
+/* mpix — MPIX_Init / MPIX_Finalize: runtime bring-up and teardown.
+ *
+ * Reference counterpart: /root/reference/src/init.cpp:157-275.  Differences:
+ *  - dual bootstrap: works inside an MPI program (MPI mode) OR from plain
+ *    torchrun-style env vars (RANK/WORLD_SIZE/MASTER_ADDR) with no MPI_Init;
+ *  - runs with zero GPUs (flag pool falls back to plain allocation) — the
+ *    reference's MPIACX_DISABLE_MEMOPS path still required CUDA;
+ *  - the memOps fast path is probed FUNCTIONALLY (a test hipStreamWriteValue32
+ *    round-trip), not just via a device attribute;
+ *  - proxy thread lives in proxy.cpp and watches only active slots.
+ */
+#include <hip/hip_runtime.h>
+
+#include "internal.h"
+
+namespace mpix {
+
+static int env_int(const char *name, int dflt)
+{
+    const char *v = getenv(name);
+    if (!v || !*v) return dflt;
+    return atoi(v);
+}
+
+/* Functional probe of hipStreamWriteValue32 / hipStreamWaitValue32 /
+ * hipStreamBatchMemOp against the pinned flag pool. */
+static void probe_memops(State *s)
+{
+    s->use_memops = false;
+    s->use_batch_memops = false;
+    if (!s->have_gpu) return;
+    if (env_int("MPIX_DISABLE_MEMOPS", 0)) {
+        MPIX_DBG("memOps disabled by MPIX_DISABLE_MEMOPS");
+        return;
+    }
+    int can = 0;
+    (void)hipDeviceGetAttribute(&can, hipDeviceAttributeCanUseStreamWaitValue,
+                                s->device_id);
+    hipStream_t st;
+    if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) != hipSuccess) {
+        (void)hipGetLastError();
+        return;
+    }
+    const int probe_idx = (int)s->nflags - 1;
+    uint32_t *fd = s->flags_d + probe_idx;
+    const uint32_t magic = 0xC0FFEEu;
+
+    do {
+        if (hipStreamWriteValue32(st, fd, magic, 0) != hipSuccess) break;
+        if (hipStreamSynchronize(st) != hipSuccess) break;
+        if (s->flags[probe_idx].load() != magic) break;
+        if (hipStreamWaitValue32(st, fd, magic, hipStreamWaitValueEq, 0xFFFFFFFFu)
+            != hipSuccess) break;
+        if (hipStreamSynchronize(st) != hipSuccess) break;
+        s->use_memops = true;
+
+        hipStreamBatchMemOpParams p[1];
+        memset(p, 0, sizeof(p));
+        p[0].writeValue.operation = hipStreamMemOpWriteValue32;
+        p[0].writeValue.address = fd;
+        p[0].writeValue.value = magic + 1;
+        if (hipStreamBatchMemOp(st, 1, p, 0) != hipSuccess) break;
+        if (hipStreamSynchronize(st) != hipSuccess) break;
+        if (s->flags[probe_idx].load() != magic + 1) break;
+        s->use_batch_memops = true;
+    } while (0);
+    (void)hipGetLastError();
+    (void)hipStreamSynchronize(st);
+    (void)hipStreamDestroy(st);
+    s->flags[probe_idx].store(MPIX_FLAG_AVAILABLE);
+    if (!s->use_memops) {
+        fprintf(stderr,
+                "[mpix] warn: hipStream memOps unavailable (attr=%d); "
+                "falling back to trigger/wait kernels\n", can);
+    }
+}
+
+extern "C" int MPIX_Init(void)
+{
+    if (g_state != nullptr) {
+        MPIX_ERR("MPIX_Init called twice");
+        return MPI_ERR_OTHER;
+    }
+    State *s = new State();
+
+    /* --- identity: MPI mode or env mode ------------------------------- */
+    int mpi_inited = 0, mpi_finalized = 0;
+    MPI_Initialized(&mpi_inited);
+    MPI_Finalized(&mpi_finalized);
+    if (mpi_finalized) {
+        MPIX_ERR("MPI already finalized");
+        delete s;
+        return MPI_ERR_OTHER;
+    }
+    if (mpi_inited) {
+        s->mpi_mode = true;
+        int provided = 0;
+        MPI_Query_thread(&provided);
+        if (provided < MPI_THREAD_MULTIPLE) {
+            MPIX_ERR("MPI mode requires MPI_THREAD_MULTIPLE "
+                     "(the proxy thread calls MPI); got %d", provided);
+            delete s;
+            return MPI_ERR_OTHER;
+        }
+        MPI_Comm_rank(MPI_COMM_WORLD, &s->world_rank);
+        MPI_Comm_size(MPI_COMM_WORLD, &s->world_size);
+    } else {
+        s->mpi_mode = false;
+        s->world_rank = env_int("RANK", 0);
+        s->world_size = env_int("WORLD_SIZE", 1);
+    }
+
+    /* --- GPU ----------------------------------------------------------- */
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess) {
+        (void)hipGetLastError();
+        ndev = 0;
+    }
+    if (ndev > 0 && !env_int("MPIX_FORCE_NO_GPU", 0)) {
+        s->have_gpu = true;
+        int local = env_int("LOCAL_RANK", s->world_rank);
+        s->device_id = local % ndev;
+        if (hipSetDevice(s->device_id) != hipSuccess) {
+            MPIX_ERR("hipSetDevice(%d) failed", s->device_id);
+            delete s;
+            return MPI_ERR_OTHER;
+        }
+    }
+
+    /* --- flag pool + op table ------------------------------------------ */
+    long nf = env_int("MPIX_NFLAGS", 4096);
+    if (nf < 64) nf = 64;
+    s->nflags = (size_t)nf;
+    void *raw = nullptr;
+    if (s->have_gpu) {
+        if (hipHostMalloc(&raw, s->nflags * sizeof(uint32_t),
+                          hipHostMallocMapped) != hipSuccess) {
+            MPIX_ERR("hipHostMalloc(flag pool) failed");
+            delete s;
+            return MPI_ERR_OTHER;
+        }
+        s->flags_pinned = true;
+        void *dptr = nullptr;
+        if (hipHostGetDevicePointer(&dptr, raw, 0) != hipSuccess) {
+            MPIX_ERR("hipHostGetDevicePointer failed");
+            delete s;
+            return MPI_ERR_OTHER;
+        }
+        s->flags_d = (uint32_t *)dptr;
+    } else {
+        raw = calloc(s->nflags, sizeof(uint32_t));
+        s->flags_d = nullptr;
+    }
+    s->flags = reinterpret_cast<std::atomic<uint32_t> *>(raw);
+    for (size_t i = 0; i < s->nflags; i++)
+        s->flags[i].store(MPIX_FLAG_AVAILABLE, std::memory_order_relaxed);
+    s->ops = new Op[s->nflags];
+    s->armed.init(4 * s->nflags);
+    s->spin_before_yield = env_int("MPIX_PROXY_SPIN", 2000);
+
+    g_state = s; /* utilities below use g_state */
+
+    /* --- memOps fast path ---------------------------------------------- */
+    probe_memops(s);
+
+    /* --- data plane ----------------------------------------------------- */
+    s->t_native = make_native_transport(s->world_rank, s->world_size,
+                                        s->mpi_mode, s->have_gpu, s->device_id);
+    if (s->t_native == nullptr) {
+        MPIX_ERR("native transport bring-up failed");
+        g_state = nullptr;
+        delete s;
+        return MPI_ERR_OTHER;
+    }
+    if (s->mpi_mode) s->t_mpi = make_mpi_transport();
+
+    /* --- proxy ----------------------------------------------------------- */
+    s->proxy_stop.store(false);
+    s->proxy = std::thread(proxy_main);
+
+    MPIX_DBG("init done: rank %d/%d mpi_mode=%d gpu=%d dev=%d memops=%d batch=%d "
+             "nflags=%zu", s->world_rank, s->world_size, s->mpi_mode,
+             s->have_gpu, s->device_id, s->use_memops, s->use_batch_memops,
+             s->nflags);
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Finalize(void)
+{
+    State *s = g_state;
+    if (s == nullptr) return MPI_ERR_OTHER;
+
+    s->proxy_stop.store(true);
+    if (s->proxy.joinable()) s->proxy.join();
+
+    size_t leaked = 0;
+    for (size_t i = 0; i < s->nflags; i++) {
+        uint32_t f = s->flags[i].load(std::memory_order_relaxed);
+        if (f != MPIX_FLAG_AVAILABLE) leaked++;
+    }
+    if (leaked)
+        fprintf(stderr, "[mpix] warn: %zu flag slot(s) still in use at "
+                "MPIX_Finalize\n", leaked);
+
+    if (s->t_native) {
+        native_transport_shutdown(s->t_native);
+        delete s->t_native;
+    }
+    delete s->t_mpi;
+
+    delete[] s->ops;
+    if (s->flags_pinned) (void)hipHostFree((void *)s->flags);
+    else free((void *)s->flags);
+
+    g_state = nullptr;
+    delete s;
+    return MPI_SUCCESS;
+}
+
+} /* namespace mpix */

@@ -1,0 +1,300 @@
+/* mpix — internal data model: state machine, op/request descriptors, globals.
+ *
+ * Design notes (vs the reference, /root/reference/include/mpi-acx-internal.h):
+ *  - flags are C++ std::atomic<uint32_t> words in host-pinned fine-grained
+ *    memory (device writes them with system-scope HIP atomics), not volatile.
+ *  - the slot allocator is lock-free CAS with a rotating cursor (fixes the
+ *    reference's single-issuer FIXME, triggered.cpp:40-44).
+ *  - the proxy watches only ACTIVE slots (an MPSC hand-off ring + a
+ *    proxy-owned watch list) instead of scanning the whole pool every pass
+ *    (reference scans all 4096 flags, init.cpp:61).
+ *  - CLEANUP is handled at the top level of the proxy loop (fixes the
+ *    reference's slot leak, D1 in SURVEY.md).
+ *  - the data plane is a pluggable Transport (native shm/xGMI or host MPI),
+ *    not a hard dependency on a GPU-aware MPI.
+ */
+#ifndef MPIX_INTERNAL_H
+#define MPIX_INTERNAL_H
+
+#include <atomic>
+#include <cstdint>
+#include <memory>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include <mpi.h>
+
+#include "mpix/mpix.h"
+#include "mpix/mpix_abi.h"
+
+namespace mpix {
+
+/* ------------------------------------------------------------------ errors */
+
+#define MPIX_ERR(...)                                                   \
+    do {                                                                \
+        fprintf(stderr, "[mpix] %s:%d %s: ", __FILE__, __LINE__, __func__); \
+        fprintf(stderr, __VA_ARGS__);                                   \
+        fprintf(stderr, "\n");                                          \
+    } while (0)
+
+#define MPIX_CHECK(call)                                                \
+    do {                                                                \
+        int _e = (call);                                                \
+        if (_e != 0) { MPIX_ERR("%s failed (%d)", #call, _e); return _e; } \
+    } while (0)
+
+#define MPIX_CHECK_HIP(call)                                            \
+    do {                                                                \
+        hipError_t _e = (call);                                         \
+        if (_e != hipSuccess) {                                         \
+            MPIX_ERR("%s failed: %s", #call, hipGetErrorString(_e));    \
+            return (int)_e;                                             \
+        }                                                               \
+    } while (0)
+
+#ifdef MPIX_DEBUG
+#define MPIX_DBG(...)                                                   \
+    do {                                                                \
+        fprintf(stderr, "[mpix dbg r%d] ", mpix::g_state ? mpix::g_state->world_rank : -1); \
+        fprintf(stderr, __VA_ARGS__);                                   \
+        fprintf(stderr, "\n");                                          \
+    } while (0)
+#else
+#define MPIX_DBG(...) do {} while (0)
+#endif
+
+/* ------------------------------------------------------------- descriptors */
+
+struct Request;
+
+/* Transport-level completion record (filled by the channel). */
+struct ChStatus {
+    int src  = -1;   /* world rank of the actual sender */
+    int tag  = -1;
+    uint64_t bytes = 0;
+    int err  = 0;
+};
+
+enum class OpKind : int32_t { NONE = 0, ISEND, IRECV, PSEND_PART, PRECV_PART };
+
+/* One in-flight operation, bound 1:1 to a flag slot. */
+struct Op {
+    OpKind kind = OpKind::NONE;
+    void *buf = nullptr;
+    uint64_t bytes = 0;
+    int count = 0;
+    MPI_Datatype datatype = MPI_DATATYPE_NULL;
+    int peer = -1;        /* rank as passed by the user (comm-relative) */
+    int peer_world = -1;  /* resolved world rank (MPI_ANY_SOURCE passes through) */
+    int tag = 0;
+    MPI_Comm comm = MPI_COMM_NULL;
+    uint32_t comm_id = 0;
+    bool buf_is_device = false;
+    bool native_route = true;       /* native shm/xGMI channel vs MPI passthrough */
+    int partition = -1;             /* >=0 for partitioned partial ops */
+    uint32_t pseq = 0;              /* partitioned iteration number */
+    Request *req = nullptr;         /* owning request */
+
+    /* transport hand-off (written by channel, read by proxy) */
+    std::atomic<int> ch_done{0};
+    ChStatus ch_status;
+    void *ch_priv = nullptr;        /* channel-private per-op state */
+
+    /* completion/status delivery (guarded by g_state->completion_mutex) */
+    MPI_Status *enq_status_target = nullptr; /* posted by MPIX_Wait*_enqueue */
+    MPI_Status saved_status;
+    bool status_saved = false;
+    bool orphaned = false;          /* user called MPIX_Request_free pre-completion */
+
+    void reset() {
+        kind = OpKind::NONE; buf = nullptr; bytes = 0; count = 0;
+        datatype = MPI_DATATYPE_NULL; peer = peer_world = -1; tag = 0;
+        comm = MPI_COMM_NULL; comm_id = 0; buf_is_device = false;
+        native_route = true;
+        partition = -1; pseq = 0; req = nullptr;
+        ch_done.store(0, std::memory_order_relaxed);
+        ch_status = ChStatus{}; ch_priv = nullptr;
+        enq_status_target = nullptr; status_saved = false; orphaned = false;
+    }
+};
+
+enum class ReqKind : int32_t { BASIC = 0, PARTITIONED };
+
+struct Request {
+    ReqKind kind = ReqKind::BASIC;
+
+    /* BASIC */
+    int flag_idx = -1;
+
+    /* PARTITIONED */
+    bool is_send = false;
+    int n_partitions = 0;
+    std::vector<int> part_idx;      /* partition -> flag slot */
+    void *buf = nullptr;
+    uint64_t part_bytes = 0;        /* bytes per partition */
+    int peer = -1, peer_world = -1, tag = 0;
+    MPI_Comm comm = MPI_COMM_NULL;
+    uint32_t comm_id = 0;
+    MPI_Datatype datatype = MPI_DATATYPE_NULL;
+    int count_per_part = 0;
+    bool active = false;            /* between Start and Wait */
+    uint32_t start_seq = 0;         /* number of MPIX_Start calls */
+    mpix_prequest_dev_t *dev_handle = nullptr; /* device copy (Prequest_create) */
+    int32_t *dev_idx = nullptr;                /* device idx array */
+};
+
+/* ----------------------------------------------------------------- channel */
+
+class Transport {
+public:
+    virtual ~Transport() = default;
+    /* Hand an ISEND/IRECV/PSEND_PART/PRECV_PART op to the data plane.
+     * Never blocks; completion is signalled via op->ch_done / op->ch_status. */
+    virtual int start(Op *op) = 0;
+    /* Advance all in-flight traffic. Called from the proxy loop. */
+    virtual void progress() = 0;
+    virtual const char *name() const = 0;
+};
+
+/* --------------------------------------------------------------- MPSC ring */
+
+/* Fixed-size MPSC ring for handing freshly armed slots to the proxy.
+ * Producers: any app thread calling an enqueue/init API. Consumer: proxy. */
+class SlotRing {
+public:
+    void init(size_t capacity) {
+        cap_ = 1; while (cap_ < capacity) cap_ <<= 1;
+        buf_.reset(new std::atomic<int64_t>[cap_]);
+        for (size_t i = 0; i < cap_; i++) buf_[i].store(EMPTY, std::memory_order_relaxed);
+        head_.store(0); tail_.store(0);
+    }
+    void push(int idx) {
+        uint64_t pos = head_.fetch_add(1, std::memory_order_relaxed);
+        /* wait for the slot to be consumed if we lapped (practically never:
+         * capacity >= 4 * nflags > max outstanding) */
+        auto &cell = buf_[pos & (cap_ - 1)];
+        int64_t expected = EMPTY;
+        while (!cell.compare_exchange_weak(expected, idx,
+                                           std::memory_order_release,
+                                           std::memory_order_relaxed)) {
+            expected = EMPTY;
+            std::this_thread::yield();
+        }
+    }
+    /* proxy-only */
+    bool pop(int *idx) {
+        uint64_t t = tail_.load(std::memory_order_relaxed);
+        auto &cell = buf_[t & (cap_ - 1)];
+        int64_t v = cell.load(std::memory_order_acquire);
+        if (v == EMPTY) return false;
+        cell.store(EMPTY, std::memory_order_relaxed);
+        tail_.store(t + 1, std::memory_order_relaxed);
+        *idx = (int)v;
+        return true;
+    }
+private:
+    static constexpr int64_t EMPTY = -1;
+    size_t cap_ = 0;
+    std::unique_ptr<std::atomic<int64_t>[]> buf_;
+    std::atomic<uint64_t> head_{0};
+    std::atomic<uint64_t> tail_{0};
+};
+
+/* ------------------------------------------------------------------- state */
+
+struct State {
+    /* identity */
+    int world_rank = 0;
+    int world_size = 1;
+    bool mpi_mode = false;      /* MPI_Init was called by the app */
+    /* gpu */
+    bool have_gpu = false;
+    int device_id = -1;
+    bool use_memops = false;    /* hipStreamWriteValue32/WaitValue32 path */
+    bool use_batch_memops = false;
+    /* flag pool */
+    size_t nflags = 0;
+    std::atomic<uint32_t> *flags = nullptr;  /* host view (pinned if GPU) */
+    uint32_t *flags_d = nullptr;             /* device view or nullptr */
+    bool flags_pinned = false;
+    Op *ops = nullptr;                       /* parallel op table */
+    std::atomic<uint32_t> alloc_cursor{0};
+    /* proxy */
+    std::thread proxy;
+    std::atomic<bool> proxy_stop{false};
+    SlotRing armed;                          /* slots newly allocated */
+    std::mutex completion_mutex;             /* status-delivery race guard */
+    /* data plane */
+    Transport *t_native = nullptr;   /* shm + xGMI/IPC intra-node channel */
+    Transport *t_mpi = nullptr;      /* host-MPI passthrough (MPI mode only) */
+    /* proxy tuning */
+    int spin_before_yield = 2000;
+    /* stats */
+    std::atomic<uint64_t> ops_issued{0};
+    std::atomic<uint64_t> ops_completed{0};
+};
+
+extern State *g_state;
+
+/* ------------------------------------------------------------- flag helpers */
+
+static inline uint32_t flag_load(int idx) {
+    return g_state->flags[idx].load(std::memory_order_acquire);
+}
+static inline void flag_store(int idx, uint32_t v) {
+    g_state->flags[idx].store(v, std::memory_order_release);
+}
+static inline bool flag_cas(int idx, uint32_t expect, uint32_t v) {
+    return g_state->flags[idx].compare_exchange_strong(
+        expect, v, std::memory_order_acq_rel, std::memory_order_relaxed);
+}
+
+/* ------------------------------------------------------------------- slots */
+
+/* Allocate a flag slot: AVAILABLE -> RESERVED (lock-free, multi-producer).
+ * Returns slot index or -1 if the pool is exhausted. */
+int slot_allocate();
+/* Release a slot back to AVAILABLE (resets the op descriptor first). */
+void slot_free(int idx);
+/* Arm a slot: make the proxy watch it (call after the op is fully described). */
+void slot_arm(int idx);
+
+/* ------------------------------------------------------------------- utils */
+
+/* Size in bytes of an MPI datatype; works without MPI_Init for MPICH builtin
+ * handles (size lives in bits 8..15 of the handle). */
+int datatype_size(MPI_Datatype dt, int *size_out);
+
+/* Resolve a comm-relative rank to a world rank; assigns/validates comm_id.
+ * comm_id 0 = WORLD, 1 = SELF; other comms are MPI-mode passthrough only. */
+int resolve_peer(MPI_Comm comm, int rank, int *world_rank_out,
+                 uint32_t *comm_id_out, bool *native_ok_out);
+
+/* Write an MPI_Status (MPICH layout) from a transport completion. */
+void fill_status(MPI_Status *st, const ChStatus &cs);
+
+/* Is `ptr` device memory? (false when no GPU) */
+bool ptr_is_device(const void *ptr);
+
+/* proxy entry point (init.cpp spawns it) */
+void proxy_main();
+
+/* enqueue.cpp internals shared with init */
+int trigger_host(int idx);   /* host-side trigger: flag -> PENDING */
+
+/* transport factories */
+Transport *make_native_transport(int world_rank, int world_size, bool mpi_mode,
+                                 bool have_gpu, int device_id);
+Transport *make_mpi_transport();
+
+/* native transport finalize hook (unlinks shm) */
+void native_transport_shutdown(Transport *t);
+
+} /* namespace mpix */
+
+#endif /* MPIX_INTERNAL_H */

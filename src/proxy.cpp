@@ -1,0 +1,192 @@
+/* mpix — the CPU proxy/progress thread.
+ *
+ * Reference counterpart: progress_thread_fn (/root/reference/src/init.cpp:55-154).
+ * Redesigned:
+ *  - watches only ACTIVE slots (armed-slot ring + watch list) instead of a
+ *    full O(nflags) pool scan per pass;
+ *  - CLEANUP handled at the TOP level of the loop (fixes reference defect D1,
+ *    where stream-waited slots leaked until finalize);
+ *  - issue/complete go through the pluggable Transport (native shm/xGMI or
+ *    host-MPI), never direct MPI calls;
+ *  - exponential idle backoff (spin -> yield) so an idle proxy does not pin
+ *    a core at 100% forever (tunable via MPIX_PROXY_SPIN).
+ *
+ * Flag lifecycle driven here:
+ *   PENDING   -> transport->start(op) -> ISSUED
+ *   ISSUED    -> op->ch_done          -> COMPLETED (+status save/delivery)
+ *   COMPLETED -> (orphaned only) free
+ *   CLEANUP   -> free slot (+request for basic ops)
+ */
+#include <hip/hip_runtime.h>
+
+#include <chrono>
+
+#include "internal.h"
+
+namespace mpix {
+
+static inline Transport *route(Op *op)
+{
+    State *s = g_state;
+    if (op->native_route || s->t_mpi == nullptr) return s->t_native;
+    return s->t_mpi;
+}
+
+/* Transition ISSUED -> COMPLETED: save/deliver status under the completion
+ * mutex (closes the proxy-vs-Wait race the reference handles with
+ * mpiacx_op_completion_mutex + try_complete_wait_op, sendrecv.cu:82-104). */
+static bool complete_op(int idx, Op *op)
+{
+    State *s = g_state;
+    std::lock_guard<std::mutex> lk(s->completion_mutex);
+    fill_status(&op->saved_status, op->ch_status);
+    op->status_saved = true;
+    if (op->enq_status_target != nullptr) {
+        *op->enq_status_target = op->saved_status;
+    }
+    s->ops_completed.fetch_add(1, std::memory_order_relaxed);
+    if (op->orphaned &&
+        (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV)) {
+        /* user already called MPIX_Request_free: nobody will wait */
+        delete op->req;
+        slot_free(idx);
+        return true; /* slot gone */
+    }
+    flag_store(idx, MPIX_FLAG_COMPLETED);
+    return false;
+}
+
+void proxy_main()
+{
+    State *s = g_state;
+    if (s->have_gpu) (void)hipSetDevice(s->device_id);
+
+    std::vector<int> watch;
+    watch.reserve(s->nflags);
+    std::vector<uint8_t> watched(s->nflags, 0);
+    int idle = 0;
+
+    auto drop = [&](size_t i) {
+        watched[watch[i]] = 0;
+        watch[i] = watch.back();
+        watch.pop_back();
+    };
+
+    while (true) {
+        bool did = false;
+
+        /* adopt newly armed slots */
+        int nidx;
+        while (s->armed.pop(&nidx)) {
+            if (!watched[nidx]) {
+                watched[nidx] = 1;
+                watch.push_back(nidx);
+            }
+            did = true;
+        }
+
+        /* drive the data plane */
+        s->t_native->progress();
+        if (s->t_mpi) s->t_mpi->progress();
+
+        /* walk active slots */
+        for (size_t i = 0; i < watch.size();) {
+            int idx = watch[i];
+            Op *op = &s->ops[idx];
+            uint32_t f = flag_load(idx);
+            switch (f) {
+            case MPIX_FLAG_AVAILABLE:
+                /* freed by a host-side wait — stop watching */
+                drop(i);
+                did = true;
+                continue;
+            case MPIX_FLAG_RESERVED:
+                break; /* described but not yet triggered */
+            case MPIX_FLAG_PENDING: {
+                int rc = route(op)->start(op);
+                if (rc == 0) {
+                    flag_store(idx, MPIX_FLAG_ISSUED);
+                    s->ops_issued.fetch_add(1, std::memory_order_relaxed);
+                    did = true;
+                } else if (rc < 0) {
+                    MPIX_ERR("transport start failed (op kind %d peer %d): %d",
+                             (int)op->kind, op->peer_world, rc);
+                    op->ch_status.err = MPI_ERR_OTHER;
+                    op->ch_done.store(1, std::memory_order_release);
+                    flag_store(idx, MPIX_FLAG_ISSUED);
+                } /* rc > 0: transient (ring full) — retry next pass */
+                break;
+            }
+            case MPIX_FLAG_ISSUED:
+                if (op->ch_done.load(std::memory_order_acquire)) {
+                    if (complete_op(idx, op)) { /* orphan: slot freed */
+                        drop(i);
+                        did = true;
+                        continue;
+                    }
+                    did = true;
+                }
+                break;
+            case MPIX_FLAG_COMPLETED: {
+                /* waiting for a waiter; handle late orphaning */
+                if (op->orphaned &&
+                    (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV)) {
+                    std::lock_guard<std::mutex> lk(s->completion_mutex);
+                    if (flag_load(idx) == MPIX_FLAG_COMPLETED && op->orphaned) {
+                        delete op->req;
+                        slot_free(idx);
+                        drop(i);
+                        did = true;
+                        continue;
+                    }
+                }
+                break;
+            }
+            case MPIX_FLAG_CLEANUP: {
+                /* waiter consumed the completion (stream wait wrote CLEANUP,
+                 * or host wait chose proxy-side free) */
+                std::lock_guard<std::mutex> lk(s->completion_mutex);
+                if (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV) {
+                    delete op->req;
+                }
+                slot_free(idx);
+                drop(i);
+                did = true;
+                continue;
+            }
+            default:
+                break;
+            }
+            i++;
+        }
+
+        if (s->proxy_stop.load(std::memory_order_acquire) && watch.empty())
+            break;
+
+        if (did) {
+            idle = 0;
+        } else {
+            idle++;
+            if (idle > s->spin_before_yield) {
+                if (s->proxy_stop.load(std::memory_order_acquire)) {
+                    /* shutting down but slots still active: keep draining,
+                     * don't burn a core */
+                    std::this_thread::sleep_for(std::chrono::microseconds(50));
+                    /* bail out if only RESERVED/COMPLETED slots remain: the
+                     * app is exiting without waiting (leak warning follows) */
+                    bool in_flight = false;
+                    for (int idx2 : watch) {
+                        uint32_t f2 = flag_load(idx2);
+                        if (f2 == MPIX_FLAG_PENDING || f2 == MPIX_FLAG_ISSUED ||
+                            f2 == MPIX_FLAG_CLEANUP) { in_flight = true; break; }
+                    }
+                    if (!in_flight) break;
+                } else {
+                    std::this_thread::yield();
+                }
+            }
+        }
+    }
+}
+
+} /* namespace mpix */

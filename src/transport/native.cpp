@@ -1,0 +1,800 @@
+/* mpix — native intra-node data plane: shared-memory control + xGMI data.
+ *
+ * The reference delegates all data movement to a CUDA-aware host MPI
+ * (SURVEY.md §2b).  On an MI355X node the fast path between GPUs is direct
+ * peer-to-peer over xGMI, so this transport implements tag-matched
+ * point-to-point natively:
+ *
+ *  control plane  per ordered rank pair (s -> r), an SPSC descriptor ring in
+ *                 a POSIX shm segment owned by r, plus a chunked staging
+ *                 buffer for host payloads.
+ *  device data    sender publishes {hipIpcMemHandle, offset}; the RECEIVER
+ *                 pulls with hipMemcpyAsync over xGMI (SDMA) on a private
+ *                 stream and acks with a DONE descriptor.  Same-process /
+ *                 same-rank transfers skip IPC and use the raw pointer.
+ *  host data      sender stages through the shm chunk ring (64 KiB chunks,
+ *                 credit-based flow control); the receiver copies chunks
+ *                 straight into the posted buffer (or an unexpected-message
+ *                 heap buffer).  Send completes when fully staged (buffered-
+ *                 send semantics), so no ack round-trip.
+ *
+ * Matching: (comm_id, src rank, tag, partitioned?, partition) with
+ * MPI_ANY_SOURCE / MPI_ANY_TAG wildcards; descriptor rings are FIFO per
+ * sender, posted receives match in post order — this preserves MPI
+ * non-overtaking per (src, tag, comm) pair, which the reference explicitly
+ * does not (README.md:173-176).
+ *
+ * Single-threaded by construction: only the proxy thread touches this object
+ * after construction.
+ */
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <hip/hip_runtime.h>
+
+#include <deque>
+#include <list>
+#include <map>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "../internal.h"
+#include "bootstrap.h"
+
+namespace mpix {
+
+/* ------------------------------------------------------------- shm layout */
+
+static constexpr uint32_t RING_SLOTS_DEFAULT = 1024;
+static constexpr uint64_t CHUNK_BYTES = 64 * 1024;
+static constexpr uint32_t STAGE_CHUNKS_DEFAULT = 64; /* 4 MiB per pair */
+
+enum DescType : uint32_t {
+    DESC_HOST_CHUNK = 1, /* one staged chunk of a host-buffer message */
+    DESC_DEV_NOTIFY = 2, /* device-buffer message advertisement */
+    DESC_DONE       = 3, /* receiver ack for DESC_DEV_NOTIFY */
+};
+
+enum DescFlags : uint32_t {
+    DESCF_PARTITIONED = 1u << 0,
+    DESCF_SELF_PTR    = 1u << 1, /* ipc field holds a raw pointer (same proc) */
+};
+
+struct alignas(64) Desc {
+    uint32_t type = 0;
+    uint32_t flags = 0;
+    uint64_t token = 0;       /* sender-side slot index */
+    int32_t src_world = -1;
+    int32_t tag = 0;
+    uint32_t comm_id = 0;
+    int32_t partition = -1;
+    uint64_t msg_bytes = 0;
+    uint64_t chunk_off = 0;   /* HOST_CHUNK: offset of this chunk in the msg */
+    uint64_t chunk_bytes = 0;
+    uint64_t stage_chunk = 0; /* HOST_CHUNK: chunk index in the stage ring */
+    uint8_t ipc[64] = {};     /* DEV_NOTIFY: hipIpcMemHandle_t (or raw ptr) */
+    uint64_t ipc_off = 0;
+    int32_t src_dev = -1;
+    uint32_t _pad[9] = {};
+};
+static_assert(sizeof(Desc) == 192, "Desc layout");
+
+struct alignas(64) InboxHdr {
+    alignas(64) std::atomic<uint64_t> head;       /* producer (peer proxy) */
+    alignas(64) std::atomic<uint64_t> tail;       /* consumer (my proxy) */
+    alignas(64) std::atomic<uint64_t> stage_tail; /* chunks released by me */
+};
+
+struct ShmGeom {
+    uint32_t ring_slots;
+    uint32_t stage_chunks;
+    size_t inbox_bytes;   /* hdr + ring + stage, 64-aligned */
+    size_t segment_bytes; /* nranks inboxes */
+};
+
+static ShmGeom shm_geometry(int nranks)
+{
+    ShmGeom g;
+    g.ring_slots = RING_SLOTS_DEFAULT;
+    if (const char *p = getenv("MPIX_SHM_RING")) g.ring_slots = atoi(p);
+    g.stage_chunks = STAGE_CHUNKS_DEFAULT;
+    if (const char *p = getenv("MPIX_SHM_STAGE_CHUNKS")) g.stage_chunks = atoi(p);
+    size_t b = sizeof(InboxHdr) + (size_t)g.ring_slots * sizeof(Desc) +
+               (size_t)g.stage_chunks * CHUNK_BYTES;
+    g.inbox_bytes = (b + 63) & ~(size_t)63;
+    g.segment_bytes = g.inbox_bytes * (size_t)nranks;
+    return g;
+}
+
+/* views into one inbox */
+struct InboxView {
+    InboxHdr *hdr;
+    Desc *ring;
+    char *stage;
+};
+
+static InboxView inbox_view(void *seg_base, const ShmGeom &g, int src)
+{
+    char *p = (char *)seg_base + (size_t)src * g.inbox_bytes;
+    InboxView v;
+    v.hdr = (InboxHdr *)p;
+    v.ring = (Desc *)(p + sizeof(InboxHdr));
+    v.stage = (char *)(v.ring + g.ring_slots);
+    return v;
+}
+
+/* -------------------------------------------------------------- transport */
+
+class NativeTransport : public Transport {
+public:
+    NativeTransport(int rank, int size, bool mpi_mode, bool have_gpu, int dev)
+        : rank_(rank), size_(size), have_gpu_(have_gpu), dev_(dev),
+          mpi_mode_(mpi_mode) {}
+
+    int init();
+    void shutdown(); /* called from MPIX_Finalize before dtor */
+    ~NativeTransport() override;
+
+    int start(Op *op) override;
+    void progress() override;
+    const char *name() const override { return "native-shm-xgmi"; }
+
+private:
+    /* ---- send side ---- */
+    struct SendState {
+        Op *op;
+        uint64_t staged = 0;   /* bytes staged so far (host path) */
+        bool notified = false; /* DEV_NOTIFY emitted */
+    };
+    /* per-destination FIFO queues (preserves non-overtaking) */
+    std::map<int, std::deque<SendState>> sendq_;
+    /* device sends awaiting DONE: token -> op */
+    std::unordered_map<uint64_t, Op *> await_done_;
+    /* pending DONE descriptors we could not emit (ring full) */
+    std::deque<std::pair<int, uint64_t>> pending_done_;
+
+    /* ---- recv side ---- */
+    struct InboundMsg {
+        Desc d;                 /* first descriptor (header info) */
+        int src;                /* sending world rank */
+        Op *op = nullptr;       /* matched recv, or null */
+        std::vector<char> heap; /* unexpected host payload buffer */
+        uint64_t got = 0;       /* host bytes received so far */
+        bool dev_copy_started = false;
+    };
+    std::list<InboundMsg> inbound_;                 /* arrival order */
+    std::unordered_map<uint64_t, InboundMsg *> inbound_by_token_;
+    std::vector<Op *> posted_recvs_;                /* post order */
+
+    struct CopyInflight {
+        Op *op;
+        hipEvent_t ev;
+        int src;
+        uint64_t token;
+        ChStatus st;
+    };
+    std::list<CopyInflight> copies_;
+
+    /* ---- shm ---- */
+    ShmGeom geom_{};
+    std::string my_seg_name_;
+    std::vector<void *> seg_;       /* seg_[r] = rank r's segment base */
+    std::vector<uint64_t> out_head_; /* producer-local head per dst */
+    std::vector<uint64_t> out_stage_head_; /* producer-local chunk cursor */
+    std::vector<uint64_t> in_tail_;  /* consumer-local tail per src */
+
+    /* ---- hip ---- */
+    hipStream_t copy_stream_ = nullptr;
+    std::vector<hipEvent_t> event_pool_;
+    std::unordered_map<std::string, void *> ipc_open_;   /* handle -> ptr */
+    std::unordered_map<const void *, std::pair<void *, hipIpcMemHandle_t>>
+        ipc_get_;                                        /* buf base -> handle */
+
+    Bootstrap *boot_ = nullptr;
+
+    int rank_, size_;
+    bool have_gpu_;
+    int dev_;
+    bool mpi_mode_;
+    bool shut_ = false;
+
+    /* helpers */
+    InboxView my_inbox(int src) { return inbox_view(seg_[rank_], geom_, src); }
+    InboxView out_box(int dst) { return inbox_view(seg_[dst], geom_, rank_); }
+
+    bool ring_has_space(int dst, uint64_t need) {
+        InboxView v = out_box(dst);
+        uint64_t tail = v.hdr->tail.load(std::memory_order_acquire);
+        return out_head_[dst] + need - tail <= geom_.ring_slots;
+    }
+    void emit_desc(int dst, const Desc &d) {
+        InboxView v = out_box(dst);
+        uint64_t h = out_head_[dst];
+        v.ring[h % geom_.ring_slots] = d;
+        v.hdr->head.store(h + 1, std::memory_order_release);
+        out_head_[dst] = h + 1;
+    }
+    uint64_t stage_free_chunks(int dst) {
+        InboxView v = out_box(dst);
+        uint64_t st = v.hdr->stage_tail.load(std::memory_order_acquire);
+        return geom_.stage_chunks - (out_stage_head_[dst] - st);
+    }
+
+    hipEvent_t get_event();
+    void put_event(hipEvent_t ev);
+
+    int progress_sends();
+    int drain_inbox(int src);
+    int progress_copies();
+    void handle_desc(int src, const Desc &d, const char *stage_base);
+    void try_match_new_inbound(InboundMsg &m);
+    void attach(InboundMsg &m, Op *op);
+    void start_dev_copy(InboundMsg &m);
+    void finish_host_recv(InboundMsg &m);
+    void deliver_chunk(InboundMsg &m, const Desc &d, const char *src_chunk);
+    void erase_inbound(InboundMsg *m);
+    int memcpy_auto(void *dst, const void *src, size_t n);
+    void *map_remote(const Desc &d);
+    bool match(const Op *op, const Desc &d, int src) const;
+    void complete_send_buffered(Op *op);
+};
+
+/* ------------------------------------------------------------------- init */
+
+int NativeTransport::init()
+{
+    boot_ = make_bootstrap(rank_, size_, mpi_mode_);
+    if (!boot_) return -1;
+
+    geom_ = shm_geometry(size_);
+
+    /* create my segment */
+    char name[96];
+    unsigned seed = (unsigned)getpid() * 2654435761u + (unsigned)rank_;
+    snprintf(name, sizeof(name), "/mpix-r%d-%d-%x", rank_, (int)getpid(),
+             rand_r(&seed));
+    my_seg_name_ = name;
+    int fd = shm_open(name, O_CREAT | O_EXCL | O_RDWR, 0600);
+    if (fd < 0) {
+        MPIX_ERR("shm_open(%s) failed", name);
+        return -1;
+    }
+    if (ftruncate(fd, (off_t)geom_.segment_bytes) != 0) {
+        MPIX_ERR("ftruncate(%zu) failed", geom_.segment_bytes);
+        close(fd);
+        return -1;
+    }
+    void *base = mmap(nullptr, geom_.segment_bytes, PROT_READ | PROT_WRITE,
+                      MAP_SHARED, fd, 0);
+    close(fd);
+    if (base == MAP_FAILED) {
+        MPIX_ERR("mmap own segment failed");
+        return -1;
+    }
+    memset(base, 0, geom_.segment_bytes);
+
+    /* exchange names, open peers */
+    struct Blob { char name[96]; };
+    Blob mine{};
+    snprintf(mine.name, sizeof(mine.name), "%s", name);
+    std::vector<Blob> all(size_);
+    if (boot_->allgather(&mine, all.data(), sizeof(Blob)) != 0) {
+        MPIX_ERR("bootstrap allgather failed");
+        return -1;
+    }
+    seg_.assign(size_, nullptr);
+    seg_[rank_] = base;
+    for (int r = 0; r < size_; r++) {
+        if (r == rank_) continue;
+        int pfd = shm_open(all[r].name, O_RDWR, 0600);
+        if (pfd < 0) {
+            MPIX_ERR("shm_open(peer %s) failed", all[r].name);
+            return -1;
+        }
+        void *pb = mmap(nullptr, geom_.segment_bytes, PROT_READ | PROT_WRITE,
+                        MAP_SHARED, pfd, 0);
+        close(pfd);
+        if (pb == MAP_FAILED) {
+            MPIX_ERR("mmap peer segment failed");
+            return -1;
+        }
+        seg_[r] = pb;
+    }
+    /* everyone mapped everything -> segments can be unlinked */
+    boot_->barrier();
+    shm_unlink(my_seg_name_.c_str());
+
+    out_head_.assign(size_, 0);
+    out_stage_head_.assign(size_, 0);
+    in_tail_.assign(size_, 0);
+
+    if (have_gpu_) {
+        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
+            hipSuccess) {
+            MPIX_ERR("copy stream create failed");
+            return -1;
+        }
+    }
+    return 0;
+}
+
+void NativeTransport::shutdown()
+{
+    if (shut_) return;
+    shut_ = true;
+    /* make sure no peer is still reading our segment */
+    if (boot_) boot_->barrier();
+    for (auto &kv : ipc_open_) (void)hipIpcCloseMemHandle(kv.second);
+    ipc_open_.clear();
+    for (hipEvent_t ev : event_pool_) (void)hipEventDestroy(ev);
+    event_pool_.clear();
+    if (copy_stream_) (void)hipStreamDestroy(copy_stream_);
+    copy_stream_ = nullptr;
+    for (int r = 0; r < (int)seg_.size(); r++)
+        if (seg_[r]) munmap(seg_[r], geom_.segment_bytes);
+    seg_.clear();
+    delete boot_;
+    boot_ = nullptr;
+}
+
+NativeTransport::~NativeTransport() { shutdown(); }
+
+/* ------------------------------------------------------------------ events */
+
+hipEvent_t NativeTransport::get_event()
+{
+    if (!event_pool_.empty()) {
+        hipEvent_t ev = event_pool_.back();
+        event_pool_.pop_back();
+        return ev;
+    }
+    hipEvent_t ev = nullptr;
+    (void)hipEventCreateWithFlags(&ev, hipEventDisableTiming);
+    return ev;
+}
+
+void NativeTransport::put_event(hipEvent_t ev) { event_pool_.push_back(ev); }
+
+/* ------------------------------------------------------------------- start */
+
+int NativeTransport::start(Op *op)
+{
+    switch (op->kind) {
+    case OpKind::ISEND:
+    case OpKind::PSEND_PART:
+        sendq_[op->peer_world].push_back(SendState{op});
+        return 0;
+    case OpKind::IRECV:
+    case OpKind::PRECV_PART: {
+        /* match against already-arrived messages first (FIFO arrival order) */
+        for (auto &m : inbound_) {
+            if (m.op == nullptr && match(op, m.d, m.src)) {
+                attach(m, op);
+                return 0;
+            }
+        }
+        posted_recvs_.push_back(op);
+        return 0;
+    }
+    default:
+        return -1;
+    }
+}
+
+bool NativeTransport::match(const Op *op, const Desc &d, int src) const
+{
+    if (op->comm_id != d.comm_id) return false;
+    bool want_part = (op->kind == OpKind::PRECV_PART);
+    bool is_part = (d.flags & DESCF_PARTITIONED) != 0;
+    if (want_part != is_part) return false;
+    if (want_part && op->partition != d.partition) return false;
+    if (op->peer_world != MPI_ANY_SOURCE && op->peer_world != src) return false;
+    if (op->tag != MPI_ANY_TAG && op->tag != d.tag) return false;
+    return true;
+}
+
+/* ---------------------------------------------------------------- progress */
+
+void NativeTransport::progress()
+{
+    progress_sends();
+    for (int s = 0; s < size_; s++) drain_inbox(s);
+    progress_copies();
+
+    /* retry queued DONE acks */
+    for (size_t i = 0; i < pending_done_.size();) {
+        auto [dst, token] = pending_done_[i];
+        if (ring_has_space(dst, 1)) {
+            Desc d;
+            d.type = DESC_DONE;
+            d.token = token;
+            d.src_world = rank_;
+            emit_desc(dst, d);
+            pending_done_.erase(pending_done_.begin() + i);
+        } else {
+            i++;
+        }
+    }
+}
+
+void NativeTransport::complete_send_buffered(Op *op)
+{
+    op->ch_status.src = (op->comm_id == 1) ? 0 : rank_;
+    op->ch_status.tag = op->tag;
+    op->ch_status.bytes = op->bytes;
+    op->ch_status.err = MPI_SUCCESS;
+    op->ch_done.store(1, std::memory_order_release);
+}
+
+int NativeTransport::progress_sends()
+{
+    for (auto &kv : sendq_) {
+        int dst = kv.first;
+        auto &q = kv.second;
+        while (!q.empty()) {
+            SendState &ss = q.front();
+            Op *op = ss.op;
+            if (op->buf_is_device) {
+                if (!ring_has_space(dst, 1)) break;
+                Desc d;
+                d.type = DESC_DEV_NOTIFY;
+                d.flags = (op->kind == OpKind::PSEND_PART) ? DESCF_PARTITIONED : 0;
+                d.token = (uint64_t)(op - g_state->ops);
+                d.src_world = rank_;
+                d.tag = op->tag;
+                d.comm_id = op->comm_id;
+                d.partition = op->partition;
+                d.msg_bytes = op->bytes;
+                d.src_dev = dev_;
+                if (dst == rank_) {
+                    d.flags |= DESCF_SELF_PTR;
+                    uint64_t p = (uint64_t)(uintptr_t)op->buf;
+                    memcpy(d.ipc, &p, 8);
+                    d.ipc_off = 0;
+                } else {
+                    /* IPC handle of the allocation base + offset */
+                    void *base = nullptr;
+                    hipError_t e = hipPointerGetAttribute(
+                        &base, HIP_POINTER_ATTRIBUTE_RANGE_START_ADDR,
+                        (hipDeviceptr_t)op->buf);
+                    if (e != hipSuccess || base == nullptr) base = op->buf;
+                    auto it = ipc_get_.find(base);
+                    if (it == ipc_get_.end()) {
+                        hipIpcMemHandle_t h;
+                        if (hipIpcGetMemHandle(&h, base) != hipSuccess) {
+                            MPIX_ERR("hipIpcGetMemHandle failed (buf %p)", base);
+                            op->ch_status.err = MPI_ERR_OTHER;
+                            op->ch_done.store(1, std::memory_order_release);
+                            q.pop_front();
+                            continue;
+                        }
+                        it = ipc_get_.emplace(base, std::make_pair(base, h)).first;
+                    }
+                    memcpy(d.ipc, &it->second.second, sizeof(hipIpcMemHandle_t));
+                    d.ipc_off = (uint64_t)((char *)op->buf - (char *)base);
+                }
+                emit_desc(dst, d);
+                await_done_[d.token] = op;
+                q.pop_front();
+                continue;
+            }
+            /* host-staged path */
+            bool stalled = false;
+            while (ss.staged < op->bytes || op->bytes == 0) {
+                if (!ring_has_space(dst, 1) || stage_free_chunks(dst) == 0) {
+                    stalled = true;
+                    break;
+                }
+                uint64_t chunk_idx = out_stage_head_[dst] % geom_.stage_chunks;
+                uint64_t n = op->bytes - ss.staged;
+                if (n > CHUNK_BYTES) n = CHUNK_BYTES;
+                InboxView v = out_box(dst);
+                if (n > 0)
+                    memcpy(v.stage + chunk_idx * CHUNK_BYTES,
+                           (const char *)op->buf + ss.staged, n);
+                Desc d;
+                d.type = DESC_HOST_CHUNK;
+                d.flags = (op->kind == OpKind::PSEND_PART) ? DESCF_PARTITIONED : 0;
+                d.token = (uint64_t)(op - g_state->ops);
+                d.src_world = rank_;
+                d.tag = op->tag;
+                d.comm_id = op->comm_id;
+                d.partition = op->partition;
+                d.msg_bytes = op->bytes;
+                d.chunk_off = ss.staged;
+                d.chunk_bytes = n;
+                d.stage_chunk = chunk_idx;
+                out_stage_head_[dst]++;
+                emit_desc(dst, d);
+                ss.staged += n;
+                if (op->bytes == 0) break; /* zero-byte message: one desc */
+            }
+            if (stalled) break; /* preserve FIFO: don't start next send */
+            complete_send_buffered(op);
+            q.pop_front();
+        }
+    }
+    return 0;
+}
+
+int NativeTransport::drain_inbox(int src)
+{
+    InboxView v = my_inbox(src);
+    uint64_t head = v.hdr->head.load(std::memory_order_acquire);
+    uint64_t tail = in_tail_[src];
+    int budget = 64;
+    while (tail < head && budget-- > 0) {
+        const Desc &d = v.ring[tail % geom_.ring_slots];
+        handle_desc(src, d, v.stage);
+        tail++;
+        v.hdr->tail.store(tail, std::memory_order_release);
+        in_tail_[src] = tail;
+    }
+    return 0;
+}
+
+void NativeTransport::handle_desc(int src, const Desc &d, const char *stage_base)
+{
+    switch (d.type) {
+    case DESC_DONE: {
+        auto it = await_done_.find(d.token);
+        if (it == await_done_.end()) {
+            MPIX_ERR("DONE for unknown token %lu", (unsigned long)d.token);
+            return;
+        }
+        Op *op = it->second;
+        await_done_.erase(it);
+        complete_send_buffered(op);
+        return;
+    }
+    case DESC_DEV_NOTIFY: {
+        inbound_.emplace_back();
+        InboundMsg &m = inbound_.back();
+        m.d = d;
+        m.src = src;
+        inbound_by_token_[((uint64_t)src << 32) ^ d.token] = &m;
+        try_match_new_inbound(m);
+        return;
+    }
+    case DESC_HOST_CHUNK: {
+        uint64_t key = ((uint64_t)src << 32) ^ d.token;
+        InboundMsg *m = nullptr;
+        if (d.chunk_off == 0) {
+            inbound_.emplace_back();
+            m = &inbound_.back();
+            m->d = d;
+            m->src = src;
+            inbound_by_token_[key] = m; /* overwrite OK: FIFO => the previous
+                                           msg with this token needs no more
+                                           chunks (see erase_inbound) */
+            try_match_new_inbound(*m);
+        } else {
+            auto it = inbound_by_token_.find(key);
+            if (it != inbound_by_token_.end()) m = it->second;
+        }
+        if (m != nullptr)
+            deliver_chunk(*m, d, stage_base + d.stage_chunk * CHUNK_BYTES);
+        else
+            MPIX_ERR("chunk for unknown msg token %lu", (unsigned long)d.token);
+        /* release the stage credit unconditionally (no credit leaks) */
+        InboxView v = my_inbox(src);
+        v.hdr->stage_tail.fetch_add(1, std::memory_order_release);
+        if (m != nullptr && m->got >= m->d.msg_bytes) finish_host_recv(*m);
+        return;
+    }
+    default:
+        MPIX_ERR("bad descriptor type %u from %d", d.type, src);
+    }
+}
+
+void NativeTransport::try_match_new_inbound(InboundMsg &m)
+{
+    for (size_t i = 0; i < posted_recvs_.size(); i++) {
+        Op *op = posted_recvs_[i];
+        if (match(op, m.d, m.src)) {
+            posted_recvs_.erase(posted_recvs_.begin() + i);
+            attach(m, op);
+            return;
+        }
+    }
+    /* unexpected: host messages buffer into heap as chunks arrive */
+    if (m.d.type == DESC_HOST_CHUNK) m.heap.resize(m.d.msg_bytes);
+}
+
+void NativeTransport::attach(InboundMsg &m, Op *op)
+{
+    op->ch_priv = &m;
+    m.op = op;
+    if (m.d.type == DESC_DEV_NOTIFY) {
+        start_dev_copy(m);
+        return;
+    }
+    /* host message: move already-buffered bytes into the user buffer */
+    if (m.got > 0 && !m.heap.empty()) {
+        uint64_t n = m.got;
+        if (n > op->bytes) n = op->bytes;
+        if (n > 0) memcpy_auto(op->buf, m.heap.data(), n);
+    }
+    m.heap.clear();
+    m.heap.shrink_to_fit();
+    if (m.got >= m.d.msg_bytes) finish_host_recv(m);
+}
+
+void NativeTransport::deliver_chunk(InboundMsg &m, const Desc &d,
+                                    const char *src_chunk)
+{
+    if (m.op != nullptr) {
+        uint64_t room = (m.op->bytes > d.chunk_off)
+                            ? m.op->bytes - d.chunk_off : 0;
+        uint64_t n = d.chunk_bytes < room ? d.chunk_bytes : room;
+        if (n > 0)
+            memcpy_auto((char *)m.op->buf + d.chunk_off, src_chunk, n);
+    } else {
+        if (m.heap.size() < d.chunk_off + d.chunk_bytes)
+            m.heap.resize(m.d.msg_bytes);
+        if (d.chunk_bytes > 0)
+            memcpy(m.heap.data() + d.chunk_off, src_chunk, d.chunk_bytes);
+    }
+    m.got += d.chunk_bytes;
+    if (m.d.msg_bytes == 0) m.got = 0; /* zero-byte msg: single empty chunk */
+}
+
+void NativeTransport::finish_host_recv(InboundMsg &m)
+{
+    if (m.op == nullptr) return; /* stays buffered until a recv posts */
+    Op *op = m.op;
+    op->ch_status.src = (op->comm_id == 1) ? 0 : m.src;
+    op->ch_status.tag = m.d.tag;
+    op->ch_status.bytes =
+        m.d.msg_bytes <= op->bytes ? m.d.msg_bytes : op->bytes;
+    op->ch_status.err =
+        m.d.msg_bytes > op->bytes ? MPI_ERR_TRUNCATE : MPI_SUCCESS;
+    op->ch_done.store(1, std::memory_order_release);
+    erase_inbound(&m);
+}
+
+void *NativeTransport::map_remote(const Desc &d)
+{
+    if (d.flags & DESCF_SELF_PTR) {
+        uint64_t p;
+        memcpy(&p, d.ipc, 8);
+        return (void *)(uintptr_t)p;
+    }
+    std::string key((const char *)d.ipc, sizeof(hipIpcMemHandle_t));
+    auto it = ipc_open_.find(key);
+    void *base = nullptr;
+    if (it != ipc_open_.end()) {
+        base = it->second;
+    } else {
+        hipIpcMemHandle_t h;
+        memcpy(&h, d.ipc, sizeof(h));
+        if (hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess) !=
+            hipSuccess) {
+            MPIX_ERR("hipIpcOpenMemHandle failed (src %d)", d.src_world);
+            return nullptr;
+        }
+        ipc_open_.emplace(std::move(key), base);
+    }
+    return (char *)base + d.ipc_off;
+}
+
+void NativeTransport::start_dev_copy(InboundMsg &m)
+{
+    Op *op = m.op;
+    void *src = map_remote(m.d);
+    ChStatus st;
+    st.src = (op->comm_id == 1) ? 0 : m.src;
+    st.tag = m.d.tag;
+    st.bytes = m.d.msg_bytes <= op->bytes ? m.d.msg_bytes : op->bytes;
+    st.err = m.d.msg_bytes > op->bytes ? MPI_ERR_TRUNCATE : MPI_SUCCESS;
+    if (src == nullptr) {
+        st.err = MPI_ERR_OTHER;
+        op->ch_status = st;
+        op->ch_done.store(1, std::memory_order_release);
+        /* still ack so the sender does not hang */
+        pending_done_.emplace_back(m.src, m.d.token);
+        erase_inbound(&m);
+        return;
+    }
+    uint64_t n = st.bytes;
+    hipError_t e = hipSuccess;
+    if (n > 0)
+        e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, copy_stream_);
+    if (e != hipSuccess) {
+        MPIX_ERR("hipMemcpyAsync(pull %lu B) failed: %s", (unsigned long)n,
+                 hipGetErrorString(e));
+        st.err = MPI_ERR_OTHER;
+        op->ch_status = st;
+        op->ch_done.store(1, std::memory_order_release);
+        pending_done_.emplace_back(m.src, m.d.token);
+        erase_inbound(&m);
+        return;
+    }
+    hipEvent_t ev = get_event();
+    (void)hipEventRecord(ev, copy_stream_);
+    copies_.push_back(CopyInflight{op, ev, m.src, m.d.token, st});
+    m.dev_copy_started = true;
+    erase_inbound(&m);
+}
+
+int NativeTransport::progress_copies()
+{
+    for (auto it = copies_.begin(); it != copies_.end();) {
+        hipError_t e = hipEventQuery(it->ev);
+        if (e == hipErrorNotReady) {
+            ++it;
+            continue;
+        }
+        if (e != hipSuccess) it->st.err = MPI_ERR_OTHER;
+        put_event(it->ev);
+        /* ack the sender, then complete the recv */
+        if (ring_has_space(it->src, 1)) {
+            Desc d;
+            d.type = DESC_DONE;
+            d.token = it->token;
+            d.src_world = rank_;
+            emit_desc(it->src, d);
+        } else {
+            pending_done_.emplace_back(it->src, it->token);
+        }
+        it->op->ch_status = it->st;
+        it->op->ch_done.store(1, std::memory_order_release);
+        it = copies_.erase(it);
+    }
+    return 0;
+}
+
+void NativeTransport::erase_inbound(InboundMsg *m)
+{
+    /* erase the token mapping only if it still points at THIS message — a
+     * later message may have legitimately reused the slot-index token */
+    auto it = inbound_by_token_.find(((uint64_t)m->src << 32) ^ m->d.token);
+    if (it != inbound_by_token_.end() && it->second == m)
+        inbound_by_token_.erase(it);
+    for (auto it = inbound_.begin(); it != inbound_.end(); ++it) {
+        if (&*it == m) {
+            inbound_.erase(it);
+            return;
+        }
+    }
+}
+
+int NativeTransport::memcpy_auto(void *dst, const void *src, size_t n)
+{
+    if (!have_gpu_) {
+        memcpy(dst, src, n);
+        return 0;
+    }
+    /* hipMemcpyDefault resolves host/device direction via unified addressing;
+     * plain memcpy when both sides are host saves the runtime call. */
+    bool dd = ptr_is_device(dst);
+    if (!dd && !ptr_is_device(src)) {
+        memcpy(dst, src, n);
+        return 0;
+    }
+    return hipMemcpy(dst, src, n, hipMemcpyDefault) == hipSuccess ? 0 : -1;
+}
+
+/* ----------------------------------------------------------------- factory */
+
+Transport *make_native_transport(int world_rank, int world_size, bool mpi_mode,
+                                 bool have_gpu, int device_id)
+{
+    auto *t = new NativeTransport(world_rank, world_size, mpi_mode, have_gpu,
+                                  device_id);
+    if (t->init() != 0) {
+        delete t;
+        return nullptr;
+    }
+    return t;
+}
+
+void native_transport_shutdown(Transport *t)
+{
+    static_cast<NativeTransport *>(t)->shutdown();
+}
+
+} /* namespace mpix */
