@@ -103,6 +103,10 @@ void proxy_main()
             case MPIX_FLAG_RESERVED:
                 break; /* described but not yet triggered */
             case MPIX_FLAG_PENDING: {
+                /* graph relaunch re-fires the same slot: clear stale
+                 * completion state from the previous iteration */
+                op->ch_done.store(0, std::memory_order_relaxed);
+                op->status_saved = false;
                 int rc = route(op)->start(op);
                 if (rc == 0) {
                     flag_store(idx, MPIX_FLAG_ISSUED);
