@@ -1,0 +1,355 @@
+/* Compute/comm overlap: per-band __device__ MPIX_Pready inside a
+ * hand-written CDNA4 MFMA bf16 GEMM (BASELINE.json config 5).
+ *
+ * Each rank computes C = A x B (bf16 in, fp32 accumulate, bf16 out) with a
+ * gfx950 MFMA kernel (v_mfma_f32_16x16x32_bf16, 128x128 tiles staged through
+ * LDS, XCD-aware workgroup swizzle) and partition-sends C to its right
+ * neighbor: C's row bands are the NPARTS partitions of a persistent
+ * MPIX_Psend.  When the LAST workgroup of a band finishes its C tile it
+ * publishes the band from inside the kernel (system-release ticket counter
+ * + MPIX_Pready), so the proxy streams finished bands over xGMI while the
+ * rest of the GEMM is still running.  The receiver MPIX_Waits and verifies
+ * a sample of C against a host fp32 reference.
+ *
+ * Reports overlapped time vs GEMM-then-send, GEMM TFLOP/s, and effective
+ * bandwidth.  Run: mpiexec -np N bench/bin/gemm_pready [M N K iters]
+ * (--check: small-size full verification of the MFMA kernel itself).
+ *
+ * Publish protocol per §6 Guideline 16 of the CDNA4 guide: plain C stores
+ * -> s_waitcnt vmcnt(0) -> __syncthreads -> lane0 system release fence +
+ * restated vmcnt wait -> relaxed system fetch_add ticket; the last arriver
+ * (acquire fence) calls MPIX_Pready, itself a system-scope release store.
+ */
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+#include <mpi.h>
+
+#include "mpix/mpix.h"
+#include "mpix/mpix_device.h"
+
+#define NPARTS 64
+
+#define CHECK(cond)                                                       \
+    do {                                                                  \
+        if (!(cond)) {                                                    \
+            fprintf(stderr, "[r%d] %s:%d FAILED: %s\n", rank, __FILE__,   \
+                    __LINE__, #cond);                                     \
+            MPI_Abort(MPI_COMM_WORLD, 1);                                 \
+        }                                                                 \
+    } while (0)
+
+#define HIP(call) CHECK((call) == hipSuccess)
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) short frag8;   /* A/B: 8 bf16 */
+typedef __attribute__((ext_vector_type(4))) float frag4f;  /* C/D: 4 fp32 */
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+/* ------------------------------------------------------------- the kernel
+ * BM=BN=128, BK=32, 256 threads = 4 waves in a 2x2 wave grid; each wave
+ * owns a 64x64 quadrant = 4x4 mfma_f32_16x16x32_bf16 accumulators.
+ * A is [M][K] row-major, B is [K][N] row-major, C is [M][N] row-major.
+ * LDS images: As[128][32] linear; Bs stored transposed as Bt[128][32]
+ * (n-major) so both A and B fragments are contiguous ds_read_b128 loads
+ * (lane l of a 16x16x32 fragment reads 8 bf16 at k = (l>>4)*8).
+ */
+#define BM 128
+#define BN 128
+#define BK 32
+
+__global__ __launch_bounds__(256)
+void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
+                      bf16 *__restrict__ C, int M, int N, int K,
+                      uint32_t *band_cnt, int blocks_per_band,
+                      void *dpreq, int publish)
+{
+    __shared__ bf16 lds[BM * BK + BN * BK];
+    bf16 *As = lds;            /* [BM][BK] */
+    bf16 *Bt = lds + BM * BK;  /* [BN][BK] (transposed B tile) */
+
+    /* XCD-aware swizzle: consecutive XCDs get consecutive tile columns so
+     * each XCD's L2 sees a contiguous band of B (guide T1, bijective). */
+    int nwg = gridDim.x;
+    int wg = blockIdx.x;
+    {
+        int q = nwg / 8, r = nwg % 8, xcd = wg % 8;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
+    }
+    int tiles_n = N / BN;
+    int tm = wg / tiles_n, tn = wg % tiles_n;
+
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int wm = wave >> 1, wn = wave & 1;          /* 2x2 wave grid */
+    int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+    frag4f acc[4][4] = {};
+
+    const bf16 *Ab = A + (size_t)tm * BM * K;
+    const bf16 *Bb = B + tn * BN;
+
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        /* stage A tile: 256 threads x 8 bf16 = 4096 per pass, 2 passes */
+        {
+            const bf16 *src = Ab + k0;
+            for (int p = 0; p < 2; p++) {
+                int idx = (p * 256 + tid) * 8;       /* element offset */
+                int row = idx / BK, col = idx % BK;
+                *(frag8 *)&As[row * BK + col] =
+                    *(const frag8 *)&src[(size_t)row * K + col];
+            }
+        }
+        /* stage B tile transposed: thread loads 8 contiguous n at fixed k,
+         * scatters to Bt[n][k] (2-byte writes; correctness-first) */
+        {
+            const bf16 *src = Bb + (size_t)k0 * N;
+            for (int p = 0; p < 2; p++) {
+                int idx = (p * 256 + tid) * 8;
+                int k = idx / BN, n = idx % BN;
+                frag8 v = *(const frag8 *)&src[(size_t)k * N + n];
+                #pragma unroll
+                for (int j = 0; j < 8; j++)
+                    Bt[(n + j) * BK + k] = ((const bf16 *)&v)[j];
+            }
+        }
+        __syncthreads();
+        /* 4x4 quadrant of 16x16 tiles, two K=32 halves... BK=32 = one K */
+        frag8 af[4], bf[4];
+        #pragma unroll
+        for (int i = 0; i < 4; i++) {
+            int row = wm * 64 + i * 16 + lrow;
+            af[i] = *(const frag8 *)&As[row * BK + lk8];
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            int col = wn * 64 + j * 16 + lrow;
+            bf[j] = *(const frag8 *)&Bt[col * BK + lk8];
+        }
+        #pragma unroll
+        for (int i = 0; i < 4; i++)
+            #pragma unroll
+            for (int j = 0; j < 4; j++)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[i], bf[j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+    /* epilogue: fp32 acc -> bf16 C.  Lane l of tile (i,j) holds C rows
+     * (l>>4)*4 + r, col l&15 (C/D map col=lane&15, row=(lane>>4)*4+reg). */
+    size_t crow0 = (size_t)tm * BM + wm * 64;
+    size_t ccol0 = (size_t)tn * BN + wn * 64;
+    #pragma unroll
+    for (int i = 0; i < 4; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            size_t col = ccol0 + j * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; r++) {
+                size_t row = crow0 + i * 16 + (lane >> 4) * 4 + r;
+                C[row * N + col] = (bf16)acc[i][j][r];
+            }
+        }
+
+    if (!publish) return;
+
+    /* ---- publish the band when this is its last finishing workgroup ---- */
+    int band = (int)(crow0 * NPARTS / M); /* band of this tile's rows */
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    __shared__ int s_last;
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t prev = __hip_atomic_fetch_add(&band_cnt[band], 1,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_SYSTEM);
+        s_last = (prev == (uint32_t)blocks_per_band - 1);
+        if (s_last) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+            MPIX_Pready(band, dpreq);
+        }
+    }
+}
+
+static void host_gemm_ref(const std::vector<float> &A,
+                          const std::vector<float> &B, std::vector<float> &C,
+                          int M, int N, int K)
+{
+    for (int i = 0; i < M; i++)
+        for (int j = 0; j < N; j++) {
+            float s = 0.f;
+            for (int k = 0; k < K; k++) s += A[i * K + k] * B[k * N + j];
+            C[i * N + j] = s;
+        }
+}
+
+int main(int argc, char **argv)
+{
+    int provided, rank, size;
+    MPI_Init_thread(&argc, &argv, MPI_THREAD_MULTIPLE, &provided);
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &size);
+
+    bool check = argc > 1 && strcmp(argv[1], "--check") == 0;
+    int M = check ? 256 : (argc > 1 ? atoi(argv[1]) : 8192);
+    int N = check ? 256 : (argc > 2 ? atoi(argv[2]) : 8192);
+    int K = check ? 512 : (argc > 3 ? atoi(argv[3]) : 8192);
+    int iters = check ? 1 : (argc > 4 ? atoi(argv[4]) : 10);
+    int warmup = check ? 0 : 3;
+
+    int ndev = 0;
+    HIP(hipGetDeviceCount(&ndev));
+    CHECK(ndev > 0);
+    HIP(hipSetDevice(rank % ndev));
+    CHECK(MPIX_Init() == 0);
+    int right = (rank + 1) % size, left = (rank - 1 + size) % size;
+
+    CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
+    int nparts = check ? M / BM : NPARTS;  /* --check: 1 band per tile row */
+    CHECK(M % nparts == 0 && (M / nparts) % BM == 0);
+    int blocks_per_band = (M / nparts / BM) * (N / BN);
+
+    size_t an = (size_t)M * K, bn = (size_t)K * N, cn = (size_t)M * N;
+    bf16 *A, *B, *C, *Crecv;
+    HIP(hipMalloc(&A, an * sizeof(bf16)));
+    HIP(hipMalloc(&B, bn * sizeof(bf16)));
+    HIP(hipMalloc(&C, cn * sizeof(bf16)));
+    HIP(hipMalloc(&Crecv, cn * sizeof(bf16)));
+    uint32_t *band_cnt;
+    HIP(hipMalloc(&band_cnt, nparts * sizeof(uint32_t)));
+
+    /* fill A/B host-side: bounded pseudo-random bf16 in [-1, 1) */
+    {
+        std::vector<bf16> h(an);
+        unsigned s = 12345 + rank;
+        for (size_t i = 0; i < an; i++) {
+            s = s * 1664525u + 1013904223u;
+            h[i] = (bf16)(((float)(s >> 8) / (float)(1 << 24)) * 2.f - 1.f);
+        }
+        HIP(hipMemcpy(A, h.data(), an * sizeof(bf16), hipMemcpyHostToDevice));
+        h.resize(bn);
+        for (size_t i = 0; i < bn; i++) {
+            s = s * 1664525u + 1013904223u;
+            /* asymmetric B (guide: catches row/col-swapped C writes) */
+            h[i] = (bf16)(((float)(s >> 8) / (float)(1 << 24)) * 2.f - 1.f);
+        }
+        HIP(hipMemcpy(B, h.data(), bn * sizeof(bf16), hipMemcpyHostToDevice));
+    }
+
+    hipStream_t st;
+    HIP(hipStreamCreate(&st));
+    int grid = (M / BM) * (N / BN);
+
+    if (check) {
+        HIP(hipMemset(band_cnt, 0, nparts * sizeof(uint32_t)));
+        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
+                           A, B, C, M, N, K, band_cnt, blocks_per_band,
+                           nullptr, 0);
+        HIP(hipStreamSynchronize(st));
+        std::vector<bf16> hA(an), hB(bn), hC(cn);
+        HIP(hipMemcpy(hA.data(), A, an * sizeof(bf16), hipMemcpyDeviceToHost));
+        HIP(hipMemcpy(hB.data(), B, bn * sizeof(bf16), hipMemcpyDeviceToHost));
+        HIP(hipMemcpy(hC.data(), C, cn * sizeof(bf16), hipMemcpyDeviceToHost));
+        std::vector<float> fA(an), fB(bn), fC(cn);
+        for (size_t i = 0; i < an; i++) fA[i] = (float)hA[i];
+        for (size_t i = 0; i < bn; i++) fB[i] = (float)hB[i];
+        host_gemm_ref(fA, fB, fC, M, N, K);
+        int bad = 0;
+        float worst = 0.f;
+        for (size_t i = 0; i < cn; i++) {
+            float got = (float)hC[i], want = fC[i];
+            float err = fabsf(got - want) / (fabsf(want) + 1.f);
+            if (err > worst) worst = err;
+            if (err > 0.05f) bad++;
+        }
+        printf("[r%d] gemm --check M=%d N=%d K=%d: %d/%zu bad, worst relerr "
+               "%.4f -> %s\n", rank, M, N, K, bad, cn, worst,
+               bad ? "FAIL" : "PASS");
+        MPIX_Finalize();
+        MPI_Finalize();
+        return bad ? 1 : 0;
+    }
+
+    MPIX_Request ps, pr;
+    int count = (int)(cn / nparts);
+    CHECK(MPIX_Psend_init(C, nparts, count, MPI_SHORT, right, 21,
+                          MPI_COMM_WORLD, MPI_INFO_NULL, &ps) == 0);
+    CHECK(MPIX_Precv_init(Crecv, nparts, count, MPI_SHORT, left, 21,
+                          MPI_COMM_WORLD, MPI_INFO_NULL, &pr) == 0);
+    MPIX_Prequest dps;
+    CHECK(MPIX_Prequest_create(ps, &dps) == 0);
+
+    auto run = [&](bool overlap) {
+        auto one = [&]() {
+            MPIX_Request act[2] = {pr, ps};
+            CHECK(MPIX_Startall(2, act) == 0);
+            HIP(hipMemsetAsync(band_cnt, 0, nparts * sizeof(uint32_t), st));
+            if (overlap) {
+                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
+                                   0, st, A, B, C, M, N, K, band_cnt,
+                                   blocks_per_band, dps, 1);
+                HIP(hipStreamSynchronize(st));
+            } else {
+                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
+                                   0, st, A, B, C, M, N, K, band_cnt,
+                                   blocks_per_band, nullptr, 0);
+                HIP(hipStreamSynchronize(st));
+                for (int p = 0; p < nparts; p++)
+                    CHECK(MPIX_Pready(p, ps) == 0);
+            }
+            CHECK(MPIX_Wait(&pr, MPI_STATUS_IGNORE) == 0);
+            CHECK(MPIX_Wait(&ps, MPI_STATUS_IGNORE) == 0);
+        };
+        for (int i = 0; i < warmup; i++) one();
+        HIP(hipDeviceSynchronize());
+        MPI_Barrier(MPI_COMM_WORLD);
+        auto t0 = std::chrono::steady_clock::now();
+        for (int i = 0; i < iters; i++) one();
+        MPI_Barrier(MPI_COMM_WORLD);
+        double dt = std::chrono::duration<double>(
+                        std::chrono::steady_clock::now() - t0).count();
+        double mx;
+        MPI_Allreduce(&dt, &mx, 1, MPI_DOUBLE, MPI_MAX, MPI_COMM_WORLD);
+        return mx / iters;
+    };
+
+    /* GEMM-only reference time (no send) for TFLOP/s */
+    auto t0 = std::chrono::steady_clock::now();
+    for (int i = 0; i < iters; i++) {
+        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
+                           A, B, C, M, N, K, band_cnt, blocks_per_band,
+                           nullptr, 0);
+    }
+    HIP(hipStreamSynchronize(st));
+    double t_gemm = std::chrono::duration<double>(
+                        std::chrono::steady_clock::now() - t0).count() / iters;
+
+    double t_overlap = run(true);
+    double t_serial = run(false);
+
+    if (rank == 0) {
+        double tf = 2.0 * M * N * K / t_gemm / 1e12;
+        printf("{\"bench\": \"gemm_pready\", \"ranks\": %d, "
+               "\"mnk\": [%d,%d,%d], \"nparts\": %d, "
+               "\"gemm_tflops\": %.1f, \"ms_gemm\": %.3f, "
+               "\"ms_overlap\": %.3f, \"ms_serial\": %.3f, "
+               "\"overlap_speedup\": %.4f, \"c_bytes\": %zu}\n",
+               size, M, N, K, nparts, tf, t_gemm * 1e3, t_overlap * 1e3,
+               t_serial * 1e3, t_serial / t_overlap, cn * sizeof(bf16));
+    }
+
+    CHECK(MPIX_Prequest_free(&dps) == 0);
+    CHECK(MPIX_Request_free(&ps) == 0);
+    CHECK(MPIX_Request_free(&pr) == 0);
+    (void)hipFree(A); (void)hipFree(B); (void)hipFree(C);
+    (void)hipFree(Crecv); (void)hipFree(band_cnt);
+    (void)hipStreamDestroy(st);
+    MPIX_Finalize();
+    MPI_Finalize();
+    return 0;
+}
