@@ -60,6 +60,10 @@ def main():
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 
     use_gpu = torch.cuda.is_available()
+    if mpix.have_gpu() and not use_gpu:
+        # never run a silently degraded CPU bench on a GPU box
+        raise RuntimeError("mpix sees a GPU but torch.cuda does not — "
+                           "HIP runtime clash (import order?)")
     if use_gpu:
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dev = "cuda"
@@ -85,7 +89,11 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         return float(t.item())
 
+    def trace(msg):
+        print(f"[bench r{rank}] {msg}", file=sys.stderr, flush=True)
+
     mpix.init()
+    trace(f"init done: {mpix.config()}")
     stream = torch.cuda.current_stream() if use_gpu else None
 
     # ------------------------------------------------ 1. ping-pong half-RTT
@@ -116,9 +124,11 @@ def main():
             rs = mpix.isend_enqueue(buf, dest=peer, tag=1, stream=stream)
             mpix.wait_enqueue(rs, stream=stream)
 
+    trace("pingpong warmup")
     for _ in range(max(args.warmup * 4, 20)):
         pp_iter()
     barrier_sync()
+    trace("pingpong timed")
     t0 = time.perf_counter()
     for _ in range(args.pp_iters):
         pp_iter()
@@ -152,9 +162,11 @@ def main():
         mpix.wait(pr)
         mpix.wait(ps)
 
+    trace(f"pingpong done: half_rtt={half_rtt_us:.2f}us; psend warmup")
     for _ in range(args.warmup):
         psend_step()
     barrier_sync()
+    trace("psend timed")
     t0 = time.perf_counter()
     for _ in range(args.steps):
         psend_step()

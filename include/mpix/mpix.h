@@ -49,6 +49,12 @@ typedef void *MPIX_Prequest;
 int MPIX_Init(void);
 int MPIX_Finalize(void);
 
+/* Runtime introspection (not in the reference API): how MPIX_Init resolved
+ * this process — GPU presence, hipStream memOps probe results, bootstrap
+ * mode.  Any pointer may be NULL. */
+int MPIX_Query_config(int *have_gpu, int *use_memops, int *use_batch_memops,
+                      int *mpi_mode, int *nflags);
+
 /* ENQUEUED OPERATIONS ******************************************************/
 
 enum {

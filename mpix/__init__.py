@@ -11,6 +11,19 @@ from __future__ import annotations
 
 import os
 
+# Load torch's bundled HIP runtime FIRST if torch is present: torch ships its
+# own libamdhip64 (same SONAME libamdhip64.so.7 as /opt/rocm's). If _C.so
+# loads the /opt/rocm runtime first, torch later loads a SECOND copy (its
+# DT_NEEDED is the unversioned "libamdhip64.so", which never matches an
+# already-loaded soname) and two HSA runtimes in one process leave
+# torch.cuda.is_available() == False on a live GPU. Importing torch first
+# means _C.so's libamdhip64.so.7 dependency resolves to torch's
+# already-loaded copy and both stacks share one runtime.
+try:  # pragma: no cover - torch is optional for pure C-API consumers
+    import torch as _torch  # noqa: F401
+except ImportError:
+    pass
+
 try:
     from . import _C  # type: ignore[attr-defined]
 except ImportError as e:  # pragma: no cover
@@ -71,6 +84,11 @@ def world():
 
 def have_gpu():
     return _C.have_gpu()
+
+
+def config():
+    """Resolved runtime config (after init): gpu, memOps probe, mode."""
+    return _C.config()
 
 
 def isend_enqueue(buf, dest, tag=0, stream=None, nbytes=None):

@@ -272,6 +272,17 @@ PYBIND11_MODULE(_C, m)
         }
         return py::make_tuple(rank, size);
     });
+    m.def("config", [] {
+        int gpu = 0, mo = 0, bmo = 0, mm = 0, nf = 0;
+        PY_CHECK(MPIX_Query_config(&gpu, &mo, &bmo, &mm, &nf));
+        py::dict d;
+        d["have_gpu"] = (bool)gpu;
+        d["use_memops"] = (bool)mo;
+        d["use_batch_memops"] = (bool)bmo;
+        d["mpi_mode"] = (bool)mm;
+        d["nflags"] = nf;
+        return d;
+    });
     m.def("have_gpu", [] {
         int n = 0;
         if (hipGetDeviceCount(&n) != hipSuccess) { (void)hipGetLastError(); n = 0; }

@@ -11,6 +11,8 @@
  */
 #include <hip/hip_runtime.h>
 
+#include <chrono>
+
 #include "internal.h"
 
 namespace mpix {
@@ -52,16 +54,47 @@ static void probe_memops(State *s)
         if (hipStreamWaitValue32(st, fd, magic, hipStreamWaitValueEq, 0xFFFFFFFFu)
             != hipSuccess) break;
         if (hipStreamSynchronize(st) != hipSuccess) break;
+
+        /* HOST-write visibility: the production pattern is proxy (CPU)
+         * stores COMPLETED -> enqueued WaitValue32 must observe it.  Probe
+         * with a deadline; unwedge via a device-side write on a second
+         * stream if the polling engine cannot see host stores. */
+        if (hipStreamWaitValue32(st, fd, magic + 1, hipStreamWaitValueEq,
+                                 0xFFFFFFFFu) != hipSuccess) break;
+        s->flags[probe_idx].store(magic + 1, std::memory_order_release);
+        {
+            bool done = false;
+            for (int i = 0; i < 2000; i++) { /* ~2 s deadline */
+                hipError_t q = hipStreamQuery(st);
+                if (q == hipSuccess) { done = true; break; }
+                if (q != hipErrorNotReady) break;
+                std::this_thread::sleep_for(std::chrono::milliseconds(1));
+            }
+            (void)hipGetLastError();
+            if (!done) {
+                fprintf(stderr, "[mpix] warn: hipStreamWaitValue32 does not "
+                        "observe host stores; disabling memOps wait path\n");
+                hipStream_t st2;
+                if (hipStreamCreateWithFlags(&st2, hipStreamNonBlocking) ==
+                    hipSuccess) {
+                    (void)hipStreamWriteValue32(st2, fd, magic + 1, 0);
+                    (void)hipStreamSynchronize(st2);
+                    (void)hipStreamDestroy(st2);
+                }
+                (void)hipStreamSynchronize(st);
+                break;
+            }
+        }
         s->use_memops = true;
 
         hipStreamBatchMemOpParams p[1];
         memset(p, 0, sizeof(p));
         p[0].writeValue.operation = hipStreamMemOpWriteValue32;
         p[0].writeValue.address = fd;
-        p[0].writeValue.value = magic + 1;
+        p[0].writeValue.value = magic + 2;
         if (hipStreamBatchMemOp(st, 1, p, 0) != hipSuccess) break;
         if (hipStreamSynchronize(st) != hipSuccess) break;
-        if (s->flags[probe_idx].load() != magic + 1) break;
+        if (s->flags[probe_idx].load() != magic + 2) break;
         s->use_batch_memops = true;
     } while (0);
     (void)hipGetLastError();
@@ -182,6 +215,20 @@ extern "C" int MPIX_Init(void)
              "nflags=%zu", s->world_rank, s->world_size, s->mpi_mode,
              s->have_gpu, s->device_id, s->use_memops, s->use_batch_memops,
              s->nflags);
+    return MPI_SUCCESS;
+}
+
+extern "C" int MPIX_Query_config(int *have_gpu, int *use_memops,
+                                 int *use_batch_memops, int *mpi_mode,
+                                 int *nflags)
+{
+    State *s = g_state;
+    if (s == nullptr) return MPI_ERR_OTHER;
+    if (have_gpu) *have_gpu = s->have_gpu;
+    if (use_memops) *use_memops = s->use_memops;
+    if (use_batch_memops) *use_batch_memops = s->use_batch_memops;
+    if (mpi_mode) *mpi_mode = s->mpi_mode;
+    if (nflags) *nflags = (int)s->nflags;
     return MPI_SUCCESS;
 }
 

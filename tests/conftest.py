@@ -14,6 +14,11 @@ import socket
 import sys
 import traceback
 
+try:  # torch's HIP runtime must load before mpix._C (see mpix/__init__.py)
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 import pytest
 
 
