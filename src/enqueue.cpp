@@ -446,6 +446,28 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         return MPI_SUCCESS;
     }
 
+    /* unbatched memOps: one (wait, write) pair per request — same engine
+     * as MPIX_Wait_enqueue; used when hipStreamBatchMemOp is unavailable
+     * (probe_memops found it non-functional on ROCm 7.x mainline) */
+    if (!capturing && s->use_memops) {
+        for (int i = 0; i < count; i++) {
+            Request *req = (Request *)reqs[i];
+            if (req == nullptr) continue;
+            if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            int idx = req->flag_idx;
+            if (!try_complete_now(idx, status_at(i))) {
+                uint32_t *flag_d = s->flags_d + idx;
+                MPIX_CHECK_HIP(hipStreamWaitValue32(
+                    stream, flag_d, (uint32_t)MPIX_FLAG_COMPLETED,
+                    hipStreamWaitValueEq, 0xFFFFFFFFu));
+                MPIX_CHECK_HIP(hipStreamWriteValue32(
+                    stream, flag_d, (uint32_t)MPIX_FLAG_CLEANUP, 0));
+            }
+            reqs[i] = MPIX_REQUEST_NULL;
+        }
+        return MPI_SUCCESS;
+    }
+
     /* kernel path: one wavefront waits on all flags in a single launch.
      * The index array rides in pinned memory freed by a host callback. */
     int32_t *idx_arr = nullptr;
