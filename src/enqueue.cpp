@@ -468,7 +468,33 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         return MPI_SUCCESS;
     }
 
-    /* kernel path: one wavefront waits on all flags in a single launch.
+    /* kernel fallback (no memOps, not capturing): one k_wait_and_set per
+     * request.  A single-wave variant polling all flags in one launch
+     * (k_waitall_and_set) exists and passes isolated visibility tests but
+     * hung in situ on ROCm 7.x (store+poll in one divergent wave over the
+     * mapped pool); it is kept for the capture path (poll-only) and can be
+     * forced for experiments with MPIX_WAITALL_KERNEL=1. */
+    static const bool force_wave =
+        [] { const char *v = getenv("MPIX_WAITALL_KERNEL"); return v && atoi(v); }();
+    if (!capturing && !force_wave) {
+        for (int i = 0; i < count; i++) {
+            Request *req = (Request *)reqs[i];
+            if (req == nullptr) continue;
+            if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            int idx = req->flag_idx;
+            if (!try_complete_now(idx, status_at(i))) {
+                hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0,
+                                   stream, s->flags_d + idx,
+                                   (uint32_t)MPIX_FLAG_COMPLETED,
+                                   (uint32_t)MPIX_FLAG_CLEANUP);
+                MPIX_CHECK_HIP(hipGetLastError());
+            }
+            reqs[i] = MPIX_REQUEST_NULL;
+        }
+        return MPI_SUCCESS;
+    }
+
+    /* single-wave path: one launch polls every flag.
      * The index array rides in pinned memory freed by a host callback. */
     int32_t *idx_arr = nullptr;
     MPIX_CHECK_HIP(hipHostMalloc((void **)&idx_arr,
