@@ -210,7 +210,8 @@ int main(int argc, char **argv)
     int right = (rank + 1) % size, left = (rank - 1 + size) % size;
 
     CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
-    int nparts = check ? M / BM : NPARTS;  /* --check: 1 band per tile row */
+    /* one band >= one tile row; clamp so small M still works */
+    int nparts = check ? M / BM : (M / BM < NPARTS ? M / BM : NPARTS);
     CHECK(M % nparts == 0 && (M / nparts) % BM == 0);
     int blocks_per_band = (M / nparts / BM) * (N / BN);
 

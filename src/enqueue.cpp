@@ -468,15 +468,39 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         return MPI_SUCCESS;
     }
 
+    /* capture path: NO allocations are legal while a stream captures
+     * (hipHostMalloc returns hipErrorStreamCaptureUnsupported), so record
+     * one poll-only wait kernel per request — flags re-cycle on relaunch,
+     * cleanup is owned by the send/recv graph user objects. */
+    if (capturing) {
+        for (int i = 0; i < count; i++) {
+            Request *req = (Request *)reqs[i];
+            if (req == nullptr) continue;
+            if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            int idx = req->flag_idx;
+            {
+                std::lock_guard<std::mutex> lk(s->completion_mutex);
+                MPI_Status *st = status_at(i);
+                if (st) s->ops[idx].enq_status_target = st;
+            }
+            hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
+                               s->flags_d + idx,
+                               (uint32_t)MPIX_FLAG_COMPLETED);
+            MPIX_CHECK_HIP(hipGetLastError());
+            reqs[i] = MPIX_REQUEST_NULL;
+        }
+        return MPI_SUCCESS;
+    }
+
     /* kernel fallback (no memOps, not capturing): one k_wait_and_set per
      * request.  A single-wave variant polling all flags in one launch
      * (k_waitall_and_set) exists and passes isolated visibility tests but
      * hung in situ on ROCm 7.x (store+poll in one divergent wave over the
-     * mapped pool); it is kept for the capture path (poll-only) and can be
-     * forced for experiments with MPIX_WAITALL_KERNEL=1. */
+     * mapped pool); it is kept and can be forced for experiments with
+     * MPIX_WAITALL_KERNEL=1. */
     static const bool force_wave =
         [] { const char *v = getenv("MPIX_WAITALL_KERNEL"); return v && atoi(v); }();
-    if (!capturing && !force_wave) {
+    if (!force_wave) {
         for (int i = 0; i < count; i++) {
             Request *req = (Request *)reqs[i];
             if (req == nullptr) continue;
@@ -509,12 +533,7 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
             return MPI_ERR_REQUEST;
         }
         int idx = req->flag_idx;
-        if (capturing) {
-            std::lock_guard<std::mutex> lk(s->completion_mutex);
-            MPI_Status *st = status_at(i);
-            if (st) s->ops[idx].enq_status_target = st;
-            idx_arr[n++] = idx;
-        } else if (try_complete_now(idx, status_at(i))) {
+        if (try_complete_now(idx, status_at(i))) {
             /* done already */
         } else {
             idx_arr[n++] = idx;
@@ -524,19 +543,15 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
     if (n > 0) {
         int32_t *idx_d = nullptr;
         MPIX_CHECK_HIP(hipHostGetDevicePointer((void **)&idx_d, idx_arr, 0));
-        uint32_t newval = capturing ? 0xFFFFFFFFu : MPIX_FLAG_CLEANUP;
         int threads = n < 64 ? 64 : ((n + 63) / 64) * 64;
         if (threads > 1024) threads = 1024;
         hipLaunchKernelGGL(k_waitall_and_set, dim3(1), dim3(threads), 0,
                            stream, s->flags_d, idx_d, n,
-                           (uint32_t)MPIX_FLAG_COMPLETED, newval);
+                           (uint32_t)MPIX_FLAG_COMPLETED,
+                           (uint32_t)MPIX_FLAG_CLEANUP);
         MPIX_CHECK_HIP(hipGetLastError());
-        if (!capturing) {
-            MPIX_CHECK_HIP(hipLaunchHostFunc(
-                stream, [](void *p) { (void)hipHostFree(p); }, idx_arr));
-        }
-        /* capturing: the pinned array must outlive graph relaunches; it is
-         * intentionally leaked to the capture (small, bounded by nflags). */
+        MPIX_CHECK_HIP(hipLaunchHostFunc(
+            stream, [](void *p) { (void)hipHostFree(p); }, idx_arr));
     } else {
         (void)hipHostFree(idx_arr);
     }
