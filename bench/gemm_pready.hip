@@ -65,7 +65,7 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 __global__ __launch_bounds__(256)
 void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
                       bf16 *__restrict__ C, int M, int N, int K,
-                      uint32_t *band_cnt, int blocks_per_band,
+                      uint32_t *band_cnt, int nparts, int blocks_per_band,
                       void *dpreq, int publish)
 {
     __shared__ bf16 lds[BM * BK + BN * BK];
@@ -158,7 +158,7 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
     if (!publish) return;
 
     /* ---- publish the band when this is its last finishing workgroup ---- */
-    int band = (int)(crow0 * NPARTS / M); /* band of this tile's rows */
+    int band = (int)(crow0 * (size_t)nparts / M); /* band of this tile's rows */
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     __shared__ int s_last;
@@ -249,7 +249,7 @@ int main(int argc, char **argv)
     if (check) {
         HIP(hipMemset(band_cnt, 0, nparts * sizeof(uint32_t)));
         hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
-                           A, B, C, M, N, K, band_cnt, blocks_per_band,
+                           A, B, C, M, N, K, band_cnt, nparts, blocks_per_band,
                            nullptr, 0);
         HIP(hipStreamSynchronize(st));
         std::vector<bf16> hA(an), hB(bn), hC(cn);
@@ -292,12 +292,12 @@ int main(int argc, char **argv)
             HIP(hipMemsetAsync(band_cnt, 0, nparts * sizeof(uint32_t), st));
             if (overlap) {
                 hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
-                                   0, st, A, B, C, M, N, K, band_cnt,
+                                   0, st, A, B, C, M, N, K, band_cnt, nparts,
                                    blocks_per_band, dps, 1);
                 HIP(hipStreamSynchronize(st));
             } else {
                 hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
-                                   0, st, A, B, C, M, N, K, band_cnt,
+                                   0, st, A, B, C, M, N, K, band_cnt, nparts,
                                    blocks_per_band, nullptr, 0);
                 HIP(hipStreamSynchronize(st));
                 for (int p = 0; p < nparts; p++)
@@ -323,7 +323,7 @@ int main(int argc, char **argv)
     auto t0 = std::chrono::steady_clock::now();
     for (int i = 0; i < iters; i++) {
         hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
-                           A, B, C, M, N, K, band_cnt, blocks_per_band,
+                           A, B, C, M, N, K, band_cnt, nparts, blocks_per_band,
                            nullptr, 0);
     }
     HIP(hipStreamSynchronize(st));
