@@ -56,6 +56,24 @@ static bool complete_op(int idx, Op *op)
     return false;
 }
 
+/* MPIX_TRACE=1: state-transition tracing to stderr (runtime-gated so the
+ * production build carries observability without a debug rebuild). */
+static bool trace_on()
+{
+    static const int v = [] {
+        const char *e = getenv("MPIX_TRACE");
+        return e ? atoi(e) : 0;
+    }();
+    return v != 0;
+}
+
+#define MPIX_TRACE_EV(fmt, ...)                                           \
+    do {                                                                  \
+        if (trace_on())                                                   \
+            fprintf(stderr, "[mpix trace r%d] " fmt "\n",                 \
+                    g_state->world_rank, ##__VA_ARGS__);                  \
+    } while (0)
+
 void proxy_main()
 {
     State *s = g_state;
@@ -109,6 +127,11 @@ void proxy_main()
                 op->status_saved = false;
                 int rc = route(op)->start(op);
                 if (rc == 0) {
+                    MPIX_TRACE_EV("slot %d %s peer=%d tag=%d part=%d bytes=%lu"
+                                  " PENDING->ISSUED", idx, (int)op->kind == 1 ?
+                                  "isend" : (int)op->kind == 2 ? "irecv" :
+                                  "part", op->peer_world, op->tag,
+                                  op->partition, (unsigned long)op->bytes);
                     flag_store(idx, MPIX_FLAG_ISSUED);
                     s->ops_issued.fetch_add(1, std::memory_order_relaxed);
                     did = true;
@@ -123,6 +146,8 @@ void proxy_main()
             }
             case MPIX_FLAG_ISSUED:
                 if (op->ch_done.load(std::memory_order_acquire)) {
+                    MPIX_TRACE_EV("slot %d ISSUED->COMPLETED err=%d", idx,
+                                  op->ch_status.err);
                     if (complete_op(idx, op)) { /* orphan: slot freed */
                         drop(i);
                         did = true;
@@ -149,6 +174,7 @@ void proxy_main()
             case MPIX_FLAG_CLEANUP: {
                 /* waiter consumed the completion (stream wait wrote CLEANUP,
                  * or host wait chose proxy-side free) */
+                MPIX_TRACE_EV("slot %d CLEANUP->free", idx);
                 std::lock_guard<std::mutex> lk(s->completion_mutex);
                 if (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV) {
                     delete op->req;
