@@ -655,7 +655,8 @@ extern "C" int MPIX_Request_free(MPIX_Request *reqp)
         if (f == MPIX_FLAG_COMPLETED) {
             flag_store(idx, MPIX_FLAG_CLEANUP); /* proxy frees slot+request */
         } else {
-            s->ops[idx].orphaned = true; /* proxy frees at completion */
+            s->ops[idx].orphaned.store(true, std::memory_order_relaxed);
+            /* proxy frees at completion */
         }
         *reqp = MPIX_REQUEST_NULL;
         return MPI_SUCCESS;

@@ -109,7 +109,10 @@ struct Op {
     MPI_Status *enq_status_target = nullptr; /* posted by MPIX_Wait*_enqueue */
     MPI_Status saved_status;
     bool status_saved = false;
-    bool orphaned = false;          /* user called MPIX_Request_free pre-completion */
+    /* user called MPIX_Request_free pre-completion.  Atomic: the proxy
+     * pre-checks it outside the completion mutex (decisions that free the
+     * slot are still made under the mutex). */
+    std::atomic<bool> orphaned{false};
 
     void reset() {
         kind = OpKind::NONE; buf = nullptr; bytes = 0; count = 0;
@@ -119,7 +122,8 @@ struct Op {
         partition = -1; pseq = 0; req = nullptr;
         ch_done.store(0, std::memory_order_relaxed);
         ch_status = ChStatus{}; ch_priv = nullptr;
-        enq_status_target = nullptr; status_saved = false; orphaned = false;
+        enq_status_target = nullptr; status_saved = false;
+        orphaned.store(false, std::memory_order_relaxed);
     }
 };
 

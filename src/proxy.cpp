@@ -45,7 +45,7 @@ static bool complete_op(int idx, Op *op)
         *op->enq_status_target = op->saved_status;
     }
     s->ops_completed.fetch_add(1, std::memory_order_relaxed);
-    if (op->orphaned &&
+    if (op->orphaned.load(std::memory_order_relaxed) &&
         (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV)) {
         /* user already called MPIX_Request_free: nobody will wait */
         delete op->req;
@@ -158,10 +158,11 @@ void proxy_main()
                 break;
             case MPIX_FLAG_COMPLETED: {
                 /* waiting for a waiter; handle late orphaning */
-                if (op->orphaned &&
+                if (op->orphaned.load(std::memory_order_relaxed) &&
                     (op->kind == OpKind::ISEND || op->kind == OpKind::IRECV)) {
                     std::lock_guard<std::mutex> lk(s->completion_mutex);
-                    if (flag_load(idx) == MPIX_FLAG_COMPLETED && op->orphaned) {
+                    if (flag_load(idx) == MPIX_FLAG_COMPLETED &&
+                        op->orphaned.load(std::memory_order_relaxed)) {
                         delete op->req;
                         slot_free(idx);
                         drop(i);
