@@ -1,0 +1,43 @@
+#!/bin/bash
+# One-shot GPU validation for an MI355X box (run via gpurun).  Writes all
+# artifacts under gpurun_out/ so they merge back to the dev container.
+# Usage: bash scripts/gpu_ci.sh [quick|full]
+set -u
+cd "$(dirname "$0")/.."
+export TMPDIR=/tmp
+mkdir -p gpurun_out
+MODE="${1:-quick}"
+PASS=0; FAIL=0
+note() { echo "=== $1 ==="; }
+step() { # step <name> <timeout_s> <cmd...>
+    local name="$1" t="$2"; shift 2
+    note "$name"
+    if timeout "$t" "$@" > "gpurun_out/${name}.log" 2>&1; then
+        echo "OK"; PASS=$((PASS+1))
+    else
+        echo "FAIL rc=$?"; FAIL=$((FAIL+1)); tail -6 "gpurun_out/${name}.log"
+    fi
+}
+
+step pytest_gpu 300 python -m pytest tests -q -m gpu
+step bench_1gpu 150 python bench.py --steps 20 --warmup 5
+tail -1 gpurun_out/bench_1gpu.log
+
+export PATH=/opt/conda/bin:$PATH
+step pingpong 120 mpiexec -np 2 bench/bin/pingpong 24 100
+step halo3d 90 mpiexec -np 2 bench/bin/halo3d 192 256 256 10
+step gemm_check 90 bench/bin/gemm_pready --check
+step gemm_overlap 120 mpiexec -np 2 bench/bin/gemm_pready 4096 4096 4096 5
+
+if [ "$MODE" = full ]; then
+    step gemm_8k 90 bench/bin/gemm_pready 8192 8192 8192 5
+    note rocprof
+    ( cd /tmp && timeout 150 rocprofv3 --kernel-trace --stats \
+        -d "$OLDPWD/gpurun_out/prof" -o ci \
+        -- python "$OLDPWD/bench.py" --steps 5 --warmup 2 --msg-mib 64 \
+           --pp-iters 100 > "$OLDPWD/gpurun_out/rocprof.log" 2>&1 ) \
+        && echo OK || echo "FAIL rc=$?"
+fi
+
+echo "gpu_ci: $PASS ok, $FAIL failed ($MODE)"
+exit "$FAIL"
