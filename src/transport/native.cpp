@@ -46,6 +46,35 @@
 
 namespace mpix {
 
+/* Peer-pull copy kernel for small/medium device payloads: a shader copy
+ * over the peer mapping dodges the SDMA engine's ~40 us submit-to-complete
+ * latency (measured: the 32 KiB..2 MiB pingpong points sat flat at ~80 us
+ * half-RTT on SDMA vs ~42 us via the runtime's blit path).  Large copies
+ * stay on hipMemcpyAsync/SDMA, which frees CUs and wins on bandwidth.
+ * 16-byte vector copies with a scalar tail; grid sized for the payload. */
+__global__ void k_pull_copy(void *__restrict__ dst, const void *__restrict__ src,
+                            size_t n)
+{
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t nv = n / 16;
+    const uint4 *s4 = (const uint4 *)src;
+    uint4 *d4 = (uint4 *)dst;
+    for (size_t v = i; v < nv; v += (size_t)gridDim.x * blockDim.x)
+        d4[v] = s4[v];
+    if (i == 0)
+        for (size_t b = nv * 16; b < n; b++)
+            ((char *)dst)[b] = ((const char *)src)[b];
+}
+
+static uint64_t copy_kernel_max()
+{
+    static const uint64_t v = [] {
+        const char *e = getenv("MPIX_COPY_KERNEL_MAX");
+        return e ? (uint64_t)atoll(e) : (uint64_t)(1 << 20);
+    }();
+    return v;
+}
+
 /* ------------------------------------------------------------- shm layout */
 
 static constexpr uint32_t RING_SLOTS_DEFAULT = 1024;
@@ -701,8 +730,17 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
     }
     uint64_t n = st.bytes;
     hipError_t e = hipSuccess;
-    if (n > 0)
+    if (n > 0 && n <= copy_kernel_max()) {
+        unsigned threads = 256;
+        unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
+        if (blocks == 0) blocks = 1;
+        if (blocks > 64) blocks = 64;
+        hipLaunchKernelGGL(k_pull_copy, dim3(blocks), dim3(threads), 0,
+                           copy_stream_, op->buf, src, (size_t)n);
+        e = hipGetLastError();
+    } else if (n > 0) {
         e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, copy_stream_);
+    }
     if (e != hipSuccess) {
         MPIX_ERR("hipMemcpyAsync(pull %lu B) failed: %s", (unsigned long)n,
                  hipGetErrorString(e));
