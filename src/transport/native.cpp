@@ -70,7 +70,7 @@ static uint64_t copy_kernel_max()
 {
     static const uint64_t v = [] {
         const char *e = getenv("MPIX_COPY_KERNEL_MAX");
-        return e ? (uint64_t)atoll(e) : (uint64_t)(1 << 20);
+        return e ? (uint64_t)atoll(e) : (uint64_t)(4 << 20);
     }();
     return v;
 }
