@@ -190,6 +190,7 @@ extern "C" int MPIX_Init(void)
     s->ops = new Op[s->nflags];
     s->armed.init(4 * s->nflags);
     s->spin_before_yield = env_int("MPIX_PROXY_SPIN", 2000);
+    s->stats = env_int("MPIX_STATS", 0) != 0;
 
     g_state = s; /* utilities below use g_state */
 
@@ -248,6 +249,23 @@ extern "C" int MPIX_Finalize(void)
     if (leaked)
         fprintf(stderr, "[mpix] warn: %zu flag slot(s) still in use at "
                 "MPIX_Finalize\n", leaked);
+
+    if (s->stats) {
+        uint64_t n = s->ops_completed.load(std::memory_order_relaxed);
+        fprintf(stderr,
+                "[mpix stats r%d] ops issued=%lu completed=%lu mean "
+                "issue->complete %.1f us\n", s->world_rank,
+                (unsigned long)s->ops_issued.load(std::memory_order_relaxed),
+                (unsigned long)n,
+                n ? (double)s->lat_sum_ns / n / 1e3 : 0.0);
+        fprintf(stderr, "[mpix stats r%d] latency histogram (log2 us):",
+                s->world_rank);
+        for (int b = 0; b < 20; b++)
+            if (s->lat_hist[b])
+                fprintf(stderr, " [%d]=%lu", b,
+                        (unsigned long)s->lat_hist[b]);
+        fprintf(stderr, "\n");
+    }
 
     if (s->t_native) {
         native_transport_shutdown(s->t_native);

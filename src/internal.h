@@ -101,6 +101,7 @@ struct Op {
     Request *req = nullptr;         /* owning request */
 
     /* transport hand-off (written by channel, read by proxy) */
+    uint64_t t_issue_ns = 0;        /* proxy-side stats (MPIX_STATS=1) */
     std::atomic<int> ch_done{0};
     ChStatus ch_status;
     void *ch_priv = nullptr;        /* channel-private per-op state */
@@ -241,6 +242,11 @@ struct State {
     /* stats */
     std::atomic<uint64_t> ops_issued{0};
     std::atomic<uint64_t> ops_completed{0};
+    /* issue->complete latency histogram, log2 us buckets 0..19 (>=0.5 ms
+     * capped), proxy-thread-only (MPIX_STATS=1) */
+    bool stats = false;
+    uint64_t lat_hist[20] = {0};
+    uint64_t lat_sum_ns = 0;
 };
 
 extern State *g_state;

@@ -21,6 +21,12 @@
 
 #include <chrono>
 
+static inline uint64_t now_ns()
+{
+    return (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
+        std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+
 #include "internal.h"
 
 namespace mpix {
@@ -133,6 +139,7 @@ void proxy_main()
                                   "part", op->peer_world, op->tag,
                                   op->partition, (unsigned long)op->bytes);
                     flag_store(idx, MPIX_FLAG_ISSUED);
+                    if (s->stats) op->t_issue_ns = now_ns();
                     s->ops_issued.fetch_add(1, std::memory_order_relaxed);
                     did = true;
                 } else if (rc < 0) {
@@ -146,6 +153,14 @@ void proxy_main()
             }
             case MPIX_FLAG_ISSUED:
                 if (op->ch_done.load(std::memory_order_acquire)) {
+                    if (s->stats && op->t_issue_ns) {
+                        uint64_t d = now_ns() - op->t_issue_ns;
+                        s->lat_sum_ns += d;
+                        uint64_t us = d / 1000;
+                        int b = 0;
+                        while (us > 1 && b < 19) { us >>= 1; b++; }
+                        s->lat_hist[b]++;
+                    }
                     MPIX_TRACE_EV("slot %d ISSUED->COMPLETED err=%d", idx,
                                   op->ch_status.err);
                     if (complete_op(idx, op)) { /* orphan: slot freed */
