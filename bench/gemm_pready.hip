@@ -61,6 +61,9 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define BM 128
 #define BN 128
 #define BK 32
+#ifndef MPIX_GEMM_VARIANT
+#define MPIX_GEMM_VARIANT 1
+#endif
 
 __global__ __launch_bounds__(256)
 void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
@@ -68,9 +71,17 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
                       uint32_t *band_cnt, int nparts, int blocks_per_band,
                       void *dpreq, int publish)
 {
-    __shared__ bf16 lds[BM * BK + BN * BK];
-    bf16 *As = lds;            /* [BM][BK] */
-    bf16 *Bt = lds + BM * BK;  /* [BN][BK] (transposed B tile) */
+/* LDS row pitch: BK+8 makes every frag ds_read_b128 bank-conflict-free
+ * (rows in a 16-lane service group land on 4 distinct 64-B bank rows
+ * instead of 1; measured 4.0e9 SQ_LDS_BANK_CONFLICT cycles at pitch BK). */
+#define LP (BK + 8)
+    __shared__ bf16 lds[BM * LP + BN * LP];
+    bf16 *As = lds;            /* [BM][LP] */
+#if MPIX_GEMM_VARIANT == 2
+    bf16 *Bs = lds + BM * LP;  /* [BK][BN+8] row-major (no transpose pass) */
+#else
+    bf16 *Bt = lds + BM * LP;  /* [BN][LP] (transposed B tile) */
+#endif
 
     /* XCD-aware swizzle: consecutive XCDs get consecutive tile columns so
      * each XCD's L2 sees a contiguous band of B (guide T1, bijective). */
@@ -100,21 +111,27 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
             for (int p = 0; p < 2; p++) {
                 int idx = (p * 256 + tid) * 8;       /* element offset */
                 int row = idx / BK, col = idx % BK;
-                *(frag8 *)&As[row * BK + col] =
+                *(frag8 *)&As[row * LP + col] =
                     *(const frag8 *)&src[(size_t)row * K + col];
             }
         }
-        /* stage B tile transposed: thread loads 8 contiguous n at fixed k,
-         * scatters to Bt[n][k] (2-byte writes; correctness-first) */
+        /* stage B tile */
         {
             const bf16 *src = Bb + (size_t)k0 * N;
             for (int p = 0; p < 2; p++) {
                 int idx = (p * 256 + tid) * 8;
                 int k = idx / BN, n = idx % BN;
+#if MPIX_GEMM_VARIANT == 2
+                /* row-major image, coalesced contiguous 16-B writes */
+                *(frag8 *)&Bs[k * (BN + 8) + n] =
+                    *(const frag8 *)&src[(size_t)k * N + n];
+#else
+                /* transpose scatter (2-byte writes) */
                 frag8 v = *(const frag8 *)&src[(size_t)k * N + n];
                 #pragma unroll
                 for (int j = 0; j < 8; j++)
-                    Bt[(n + j) * BK + k] = ((const bf16 *)&v)[j];
+                    Bt[(n + j) * LP + k] = ((const bf16 *)&v)[j];
+#endif
             }
         }
         __syncthreads();
@@ -123,12 +140,19 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
         #pragma unroll
         for (int i = 0; i < 4; i++) {
             int row = wm * 64 + i * 16 + lrow;
-            af[i] = *(const frag8 *)&As[row * BK + lk8];
+            af[i] = *(const frag8 *)&As[row * LP + lk8];
         }
         #pragma unroll
         for (int j = 0; j < 4; j++) {
             int col = wn * 64 + j * 16 + lrow;
-            bf[j] = *(const frag8 *)&Bt[col * BK + lk8];
+#if MPIX_GEMM_VARIANT == 2
+            /* gather 8 ks of one column from the row-major image */
+            #pragma unroll
+            for (int e = 0; e < 8; e++)
+                ((bf16 *)&bf[j])[e] = Bs[(lk8 + e) * (BN + 8) + col];
+#else
+            bf[j] = *(const frag8 *)&Bt[col * LP + lk8];
+#endif
         }
         #pragma unroll
         for (int i = 0; i < 4; i++)
