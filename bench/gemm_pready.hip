@@ -62,7 +62,7 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define BN 128
 #define BK 32
 #ifndef MPIX_GEMM_VARIANT
-#define MPIX_GEMM_VARIANT 1
+#define MPIX_GEMM_VARIANT 2
 #endif
 
 __global__ __launch_bounds__(256)
