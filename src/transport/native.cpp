@@ -70,7 +70,7 @@ static uint64_t copy_kernel_max()
 {
     static const uint64_t v = [] {
         const char *e = getenv("MPIX_COPY_KERNEL_MAX");
-        return e ? (uint64_t)atoll(e) : (uint64_t)(4 << 20);
+        return e ? (uint64_t)atoll(e) : (uint64_t)(2 << 20);
     }();
     return v;
 }
@@ -734,7 +734,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
         unsigned threads = 256;
         unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
         if (blocks == 0) blocks = 1;
-        if (blocks > 64) blocks = 64;
+        if (blocks > 128) blocks = 128;
         hipLaunchKernelGGL(k_pull_copy, dim3(blocks), dim3(threads), 0,
                            copy_stream_, op->buf, src, (size_t)n);
         e = hipGetLastError();
