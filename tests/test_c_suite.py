@@ -13,7 +13,8 @@ TESTDIR = os.path.join(REPO, "test")
 MPIEXEC = "/opt/conda/bin/mpiexec"
 
 ALL_TESTS = ["ring", "ring_all", "ring_all_device", "ring_all_graph",
-             "ring_all_graph_construction", "ring_partitioned"]
+             "ring_all_graph_construction", "ring_partitioned",
+             "ring_subcomm"]
 
 
 def _ensure_built():
@@ -44,7 +45,8 @@ def test_c_suite_2rank(binary):
     assert ("PASS" in out) or ("SKIP" in out)
 
 
-@pytest.mark.parametrize("binary", ["ring", "ring_all", "ring_partitioned"])
+@pytest.mark.parametrize("binary", ["ring", "ring_all", "ring_partitioned",
+                                    "ring_subcomm"])
 def test_c_suite_4rank(binary):
     _ensure_built()
     assert "PASS" in _run(binary, 4)
