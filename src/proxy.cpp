@@ -90,10 +90,13 @@ void proxy_main()
     std::vector<uint8_t> watched(s->nflags, 0);
     int idle = 0;
 
+    /* Order-preserving removal: the walk below must see slots in ARM order,
+     * because two sends to one peer that both read PENDING in the same pass
+     * are issued in walk order — swap-remove compaction reordered the list
+     * and broke per-(src,tag) FIFO (caught by tests/test_soak.py). */
     auto drop = [&](size_t i) {
         watched[watch[i]] = 0;
-        watch[i] = watch.back();
-        watch.pop_back();
+        watch.erase(watch.begin() + (long)i);
     };
 
     while (true) {
