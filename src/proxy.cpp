@@ -230,8 +230,16 @@ void proxy_main()
                             f2 == MPIX_FLAG_CLEANUP) { in_flight = true; break; }
                     }
                     if (!in_flight) break;
-                } else {
+                } else if (idle < s->spin_before_yield + 512) {
                     std::this_thread::yield();
+                } else if (idle < s->spin_before_yield + 8192) {
+                    /* short naps: worst-case +10 us on a cold trigger */
+                    std::this_thread::sleep_for(std::chrono::microseconds(10));
+                } else {
+                    /* deep idle: cap the wake-up cost at 200 us; an active
+                     * stream of ops never reaches here (any progress resets
+                     * `idle`), so hot-path latency is unaffected */
+                    std::this_thread::sleep_for(std::chrono::microseconds(200));
                 }
             }
         }
