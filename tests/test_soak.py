@@ -101,3 +101,58 @@ def _soak_pair(rank, size):
 
 def test_soak_2rank():
     run_ranks(2, _soak_pair, timeout=240)
+
+
+def _soak_partitioned(rank, size):
+    """Randomized partitioned traffic: random partition counts and sizes,
+    Pready published in random order, persistent reuse over iterations,
+    interleaved with plain sends on overlapping tags."""
+    import mpix
+    mpix.init()
+    try:
+        # geometry must agree across ranks -> shared seed; per-rank rng
+        # only randomizes publish/poll ORDER
+        geom = np.random.default_rng(500)
+        rng = np.random.default_rng(900 + rank)
+        peer = (rank + 1) % size
+        left = (rank - 1 + size) % size
+        for round_ in range(6):
+            parts = int(geom.choice([1, 3, 8, 32]))
+            per = int(geom.choice([4, 100, 5000]))
+            sbuf = np.zeros(parts * per, dtype=np.int32)
+            rbuf = np.zeros(parts * per, dtype=np.int32)
+            ps = mpix.psend_init(sbuf, parts, dest=peer, tag=3)
+            pr = mpix.precv_init(rbuf, parts, source=left, tag=3)
+            for it in range(4):
+                base = rank * 1000000 + round_ * 10000 + it * 100
+                lbase = left * 1000000 + round_ * 10000 + it * 100
+                mpix.start(pr)
+                mpix.start(ps)
+                # a plain same-tag message must not confuse partitioned match
+                extra = np.full(16, base + 77, dtype=np.int32)
+                xr = mpix.isend_enqueue(extra, dest=peer, tag=3)
+                order = rng.permutation(parts)
+                for p in order:
+                    sbuf[p * per:(p + 1) * per] = base + p
+                    mpix.pready(int(p), ps)
+                # poll arrivals in a different random order
+                for p in rng.permutation(parts):
+                    while not mpix.parrived(pr, int(p)):
+                        pass
+                    seg = rbuf[p * per:(p + 1) * per]
+                    assert (seg == lbase + p).all(), f"part {p} round {round_}"
+                xbuf = np.zeros(16, dtype=np.int32)
+                rr = mpix.irecv_enqueue(xbuf, source=left, tag=3)
+                mpix.wait(rr)
+                assert (xbuf == lbase + 77).all()
+                mpix.wait(xr)
+                mpix.wait(pr)
+                mpix.wait(ps)
+            mpix.request_free(ps)
+            mpix.request_free(pr)
+    finally:
+        mpix.finalize()
+
+
+def test_soak_partitioned_2rank():
+    run_ranks(2, _soak_partitioned, timeout=300)
