@@ -66,6 +66,19 @@ __global__ void k_pull_copy(void *__restrict__ dst, const void *__restrict__ src
             ((char *)dst)[b] = ((const char *)src)[b];
 }
 
+/* Sender-push threshold for small DEVICE payloads: stage D2H into the shm
+ * chunk ring like a host message (receiver H2D's it out), skipping the
+ * NOTIFY/pull/ACK round trip entirely.  Default off (0) until measured on
+ * hardware — flip with MPIX_DEV_PUSH_MAX=<bytes> (round-2 roadmap item 1). */
+static uint64_t dev_push_max()
+{
+    static const uint64_t v = [] {
+        const char *e = getenv("MPIX_DEV_PUSH_MAX");
+        return e ? (uint64_t)atoll(e) : (uint64_t)0;
+    }();
+    return v;
+}
+
 static uint64_t copy_kernel_max()
 {
     static const uint64_t v = [] {
@@ -466,7 +479,7 @@ int NativeTransport::progress_sends()
         while (!q.empty()) {
             SendState &ss = q.front();
             Op *op = ss.op;
-            if (op->buf_is_device) {
+            if (op->buf_is_device && op->bytes > dev_push_max()) {
                 if (!ring_has_space(dst, 1)) break;
                 Desc d;
                 d.type = DESC_DEV_NOTIFY;
@@ -521,9 +534,14 @@ int NativeTransport::progress_sends()
                 uint64_t n = op->bytes - ss.staged;
                 if (n > CHUNK_BYTES) n = CHUNK_BYTES;
                 InboxView v = out_box(dst);
-                if (n > 0)
-                    memcpy(v.stage + chunk_idx * CHUNK_BYTES,
-                           (const char *)op->buf + ss.staged, n);
+                if (n > 0) {
+                    char *stage_dst = v.stage + chunk_idx * CHUNK_BYTES;
+                    const char *payload = (const char *)op->buf + ss.staged;
+                    if (op->buf_is_device)
+                        memcpy_auto(stage_dst, payload, n); /* D2H */
+                    else
+                        memcpy(stage_dst, payload, n);
+                }
                 Desc d;
                 d.type = DESC_HOST_CHUNK;
                 d.flags = (op->kind == OpKind::PSEND_PART) ? DESCF_PARTITIONED : 0;
