@@ -825,13 +825,22 @@ int NativeTransport::memcpy_auto(void *dst, const void *src, size_t n)
         return 0;
     }
     /* hipMemcpyDefault resolves host/device direction via unified addressing;
-     * plain memcpy when both sides are host saves the runtime call. */
+     * plain memcpy when both sides are host saves the runtime call.
+     * Device copies MUST ride the proxy-private non-blocking stream: a
+     * synchronous hipMemcpy runs on the legacy NULL stream, which
+     * serializes against every blocking user stream — including one parked
+     * on hipStreamWaitValue32 waiting for THIS proxy to make progress
+     * (deadlock observed with MPIX_DEV_PUSH_MAX; the reference documents
+     * the same hazard class, README.md:140-150). */
     bool dd = ptr_is_device(dst);
     if (!dd && !ptr_is_device(src)) {
         memcpy(dst, src, n);
         return 0;
     }
-    return hipMemcpy(dst, src, n, hipMemcpyDefault) == hipSuccess ? 0 : -1;
+    if (hipMemcpyAsync(dst, src, n, hipMemcpyDefault, copy_stream_) !=
+        hipSuccess)
+        return -1;
+    return hipStreamSynchronize(copy_stream_) == hipSuccess ? 0 : -1;
 }
 
 /* ----------------------------------------------------------------- factory */
