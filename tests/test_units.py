@@ -160,3 +160,58 @@ def test_status_fields_any_source(mpix_env):
     assert st["tag"] == 42
     assert st["count_bytes"] == 64
     assert (dst == src).all()
+
+
+def test_zero_count_send_recv(mpix_env):
+    mpix = mpix_env
+    src = np.zeros(0, dtype=np.int32)
+    dst = np.zeros(0, dtype=np.int32)
+    rs = mpix.isend_enqueue(src, dest=0, tag=60)
+    rr = mpix.irecv_enqueue(dst, source=0, tag=60)
+    st = mpix.wait(rr)
+    mpix.wait(rs)
+    assert st["count_bytes"] == 0
+    assert st["tag"] == 60
+
+
+def test_truncation_sets_error(mpix_env):
+    """Receiving into a smaller buffer truncates and reports MPI_ERR_TRUNCATE
+    (nonzero error), without corrupting adjacent memory."""
+    mpix = mpix_env
+    src = np.arange(100, dtype=np.int32)
+    dst = np.full(60, -1, dtype=np.int32)
+    guard = np.full(8, 123, dtype=np.int32)
+    rs = mpix.isend_enqueue(src, dest=0, tag=61)
+    rr = mpix.irecv_enqueue(dst, source=0, tag=61, nbytes=40 * 4)
+    st = mpix.wait(rr)
+    mpix.wait(rs)
+    assert st["error"] != 0
+    assert (dst[:40] == src[:40]).all()
+    assert (dst[40:] == -1).all()
+    assert (guard == 123).all()
+
+
+def test_wait_enqueue_on_partitioned_rejected(mpix_env):
+    mpix = mpix_env
+    buf = np.zeros(64, dtype=np.int32)
+    ps = mpix.psend_init(buf, 4, dest=0, tag=62)
+    with pytest.raises(RuntimeError):
+        mpix.wait_enqueue(ps)
+    # clean completion
+    pr = mpix.precv_init(np.zeros(64, dtype=np.int32), 4, source=0, tag=62)
+    mpix.start(pr)
+    mpix.start(ps)
+    for p in range(4):
+        mpix.pready(p, ps)
+    mpix.wait(pr)
+    mpix.wait(ps)
+    mpix.request_free(ps)
+    mpix.request_free(pr)
+
+
+def test_query_config_shape(mpix_env):
+    cfg = mpix_env.config()
+    assert set(cfg) == {"have_gpu", "use_memops", "use_batch_memops",
+                        "mpi_mode", "nflags"}
+    assert cfg["nflags"] >= 64
+    assert cfg["mpi_mode"] is False
