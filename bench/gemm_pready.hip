@@ -65,6 +65,143 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define MPIX_GEMM_VARIANT 2
 #endif
 
+#if MPIX_GEMM_VARIANT == 3
+/* Variant 3 (round-2 candidate, compile-validated; measure with --check
+ * first): BK=64 (half the barriers of v1/v2), double-buffered LDS, and a
+ * T14-style register prefetch — tile t+1 streams HBM->VGPRs while tile t
+ * computes, then drains into the other LDS buffer; ONE barrier per K-tile.
+ * Images stay the conflict-free padded row-major layout measured in v2. */
+#undef BK
+#define BK 64
+#define LPA (BK + 8)   /* A row pitch, 144 B (16-B aligned frag reads) */
+#define LPB (BN + 8)   /* B row pitch, 272 B */
+
+__global__ __launch_bounds__(256)
+void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
+                      bf16 *__restrict__ C, int M, int N, int K,
+                      uint32_t *band_cnt, int nparts, int blocks_per_band,
+                      void *dpreq, int publish)
+{
+    __shared__ bf16 lds[2 * (BM * LPA + BK * LPB)];
+    /* buffer b: A at b*(BM*LPA+BK*LPB), B right after its A */
+    auto As = [&](int b) -> bf16 * { return lds + b * (BM * LPA + BK * LPB); };
+    auto Bs = [&](int b) -> bf16 * { return As(b) + BM * LPA; };
+
+    int nwg = gridDim.x;
+    int wg = blockIdx.x;
+    {
+        int q = nwg / 8, r = nwg % 8, xcd = wg % 8;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
+    }
+    int tiles_n = N / BN;
+    int tm = wg / tiles_n, tn = wg % tiles_n;
+
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int wm = wave >> 1, wn = wave & 1;
+    int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+    frag4f acc[4][4] = {};
+    const bf16 *Ab = A + (size_t)tm * BM * K;
+    const bf16 *Bb = B + tn * BN;
+
+    /* per-thread slices: 4 x 16 B of A and of B per K-tile */
+    int a_row[4], a_col[4], b_k[4], b_n[4];
+    #pragma unroll
+    for (int p = 0; p < 4; p++) {
+        int ia = (p * 256 + tid) * 8;
+        a_row[p] = ia / BK; a_col[p] = ia % BK;
+        b_k[p] = ia / BN;  b_n[p] = ia % BN;
+    }
+    frag8 pa[4], pb[4];
+
+    auto load_tile = [&](int t) {
+        const bf16 *sa = Ab + (size_t)t * BK;
+        const bf16 *sb = Bb + (size_t)t * BK * N;
+        #pragma unroll
+        for (int p = 0; p < 4; p++) {
+            pa[p] = *(const frag8 *)&sa[(size_t)a_row[p] * K + a_col[p]];
+            pb[p] = *(const frag8 *)&sb[(size_t)b_k[p] * N + b_n[p]];
+        }
+    };
+    auto store_tile = [&](int buf) {
+        #pragma unroll
+        for (int p = 0; p < 4; p++) {
+            *(frag8 *)&As(buf)[a_row[p] * LPA + a_col[p]] = pa[p];
+            *(frag8 *)&Bs(buf)[b_k[p] * LPB + b_n[p]] = pb[p];
+        }
+    };
+    auto compute = [&](int buf) {
+        #pragma unroll
+        for (int kh = 0; kh < 2; kh++) {
+            frag8 af[4], bf[4];
+            int kb = kh * 32 + lk8;
+            #pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm * 64 + i * 16 + lrow;
+                af[i] = *(const frag8 *)&As(buf)[row * LPA + kb];
+            }
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int col = wn * 64 + j * 16 + lrow;
+                #pragma unroll
+                for (int e = 0; e < 8; e++)
+                    ((bf16 *)&bf[j])[e] = Bs(buf)[(kb + e) * LPB + col];
+            }
+            #pragma unroll
+            for (int i = 0; i < 4; i++)
+                #pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    };
+
+    int tiles = K / BK;
+    load_tile(0);
+    store_tile(0);
+    __syncthreads();
+    for (int t = 0; t < tiles; t++) {
+        int cur = t & 1;
+        bool more = (t + 1 < tiles);
+        if (more) load_tile(t + 1);   /* HBM -> regs, overlaps the MFMAs */
+        compute(cur);
+        if (more) store_tile(1 - cur);
+        __syncthreads();
+    }
+
+    size_t crow0 = (size_t)tm * BM + wm * 64;
+    size_t ccol0 = (size_t)tn * BN + wn * 64;
+    #pragma unroll
+    for (int i = 0; i < 4; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            size_t col = ccol0 + j * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; r++) {
+                size_t row = crow0 + i * 16 + (lane >> 4) * 4 + r;
+                C[row * N + col] = (bf16)acc[i][j][r];
+            }
+        }
+
+    if (!publish) return;
+    int band = (int)(crow0 * (size_t)nparts / M);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t prev = __hip_atomic_fetch_add(&band_cnt[band], 1,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == (uint32_t)blocks_per_band - 1) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+            MPIX_Pready(band, dpreq);
+        }
+    }
+}
+
+#else /* variants 1 and 2 */
 __global__ __launch_bounds__(256)
 void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
                       bf16 *__restrict__ C, int M, int N, int K,
@@ -199,6 +336,8 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
         }
     }
 }
+
+#endif /* MPIX_GEMM_VARIANT */
 
 static void host_gemm_ref(const std::vector<float> &A,
                           const std::vector<float> &B, std::vector<float> &C,
