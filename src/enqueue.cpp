@@ -40,6 +40,14 @@ __global__ void k_wait_flag(uint32_t *flag, uint32_t val)
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
 }
 
+__global__ void k_wait_flag_gte(uint32_t *flag, uint32_t val)
+{
+    while (__hip_atomic_load(flag, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM) < val)
+        __builtin_amdgcn_s_sleep(16);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+}
+
 __global__ void k_wait_and_set(uint32_t *flag, uint32_t val, uint32_t newval)
 {
     while (__hip_atomic_load(flag, __ATOMIC_RELAXED,
@@ -229,6 +237,19 @@ static int enqueue_sendrecv(bool is_send, void *buf, int count,
     req->flag_idx = idx;
     op->req = req;
 
+    /* fast-wait protocol: stream queues only — graphs (and captures) replay
+     * with a fixed wait value, which needs the classic EQ state machine */
+    if (s->fast_wait && qtype == MPIX_QUEUE_HIP_STREAM) {
+        bool capturing = false;
+        if (s->have_gpu && queue != nullptr)
+            capturing = stream_capturing(*(hipStream_t *)queue);
+        if (!capturing) {
+            req->fast = true;
+            req->seq = ++s->slot_seq[idx];
+            op->fast = true;
+        }
+    }
+
     slot_arm(idx);
 
     int rc = fire_trigger(idx, qtype, queue, req);
@@ -295,6 +316,37 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
     }
     int idx = req->flag_idx;
     uint32_t *flag_d = s->have_gpu ? s->flags_d + idx : nullptr;
+
+    if (req->fast && qtype == MPIX_QUEUE_HIP_STREAM) {
+        if (!s->have_gpu) return MPIX_Wait(reqp, status);
+        hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
+        {
+            std::lock_guard<std::mutex> lk(s->completion_mutex);
+            if (seq_load(idx) >= req->seq) {
+                /* already completed; slot is long gone, status is in req */
+                if (status != nullptr && status != MPI_STATUS_IGNORE)
+                    *status = req->fast_status;
+                delete req;
+                *reqp = MPIX_REQUEST_NULL;
+                return MPI_SUCCESS;
+            }
+            req->consume = 1; /* proxy frees req at completion */
+            if (status != nullptr && status != MPI_STATUS_IGNORE)
+                s->ops[idx].enq_status_target = status;
+        }
+        uint32_t *seq_d = s->seqs_d + idx;
+        if (s->use_memops) {
+            MPIX_CHECK_HIP(hipStreamWaitValue32(stream, seq_d, req->seq,
+                                                hipStreamWaitValueGte,
+                                                0xFFFFFFFFu));
+        } else {
+            hipLaunchKernelGGL(k_wait_flag_gte, dim3(1), dim3(1), 0, stream,
+                               seq_d, req->seq);
+            MPIX_CHECK_HIP(hipGetLastError());
+        }
+        *reqp = MPIX_REQUEST_NULL;
+        return MPI_SUCCESS;
+    }
 
     if (qtype == MPIX_QUEUE_HIP_GRAPH) {
         if (!s->have_gpu) return MPI_ERR_OTHER;
@@ -455,6 +507,13 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
             if (req == nullptr) continue;
             if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
             int idx = req->flag_idx;
+            if (req->fast) {
+                MPIX_Request r1 = (MPIX_Request)req;
+                MPIX_CHECK(MPIX_Wait_enqueue(&r1, status_at(i),
+                                             MPIX_QUEUE_HIP_STREAM, queue));
+                reqs[i] = MPIX_REQUEST_NULL;
+                continue;
+            }
             if (!try_complete_now(idx, status_at(i))) {
                 uint32_t *flag_d = s->flags_d + idx;
                 MPIX_CHECK_HIP(hipStreamWaitValue32(
@@ -506,6 +565,13 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
             if (req == nullptr) continue;
             if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
             int idx = req->flag_idx;
+            if (req->fast) {
+                MPIX_Request r1 = (MPIX_Request)req;
+                MPIX_CHECK(MPIX_Wait_enqueue(&r1, status_at(i),
+                                             MPIX_QUEUE_HIP_STREAM, queue));
+                reqs[i] = MPIX_REQUEST_NULL;
+                continue;
+            }
             if (!try_complete_now(idx, status_at(i))) {
                 hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0,
                                    stream, s->flags_d + idx,
@@ -564,6 +630,24 @@ static int host_wait_basic(Request *req, MPI_Status *status)
 {
     State *s = g_state;
     int idx = req->flag_idx;
+    if (req->fast) {
+        {
+            std::lock_guard<std::mutex> lk(s->completion_mutex);
+            if (seq_load(idx) < req->seq)
+                req->consume = 2; /* claim: proxy must not free req */
+        }
+        int fspins = 0;
+        while (seq_load(idx) < req->seq) {
+            if (++fspins > 4096) {
+                std::this_thread::yield();
+                fspins = 0;
+            }
+        }
+        if (status != nullptr && status != MPI_STATUS_IGNORE)
+            *status = req->fast_status;
+        delete req;
+        return MPI_SUCCESS;
+    }
     int spins = 0;
     while (flag_load(idx) != MPIX_FLAG_COMPLETED) {
         if (++spins > 4096) {
@@ -647,6 +731,17 @@ extern "C" int MPIX_Request_free(MPIX_Request *reqp)
     if (s == nullptr || reqp == nullptr) return MPI_ERR_ARG;
     Request *req = (Request *)*reqp;
     if (req == nullptr) return MPI_SUCCESS;
+
+    if (req->kind == ReqKind::BASIC && req->fast) {
+        int idx = req->flag_idx;
+        std::lock_guard<std::mutex> lk(s->completion_mutex);
+        if (seq_load(idx) >= req->seq)
+            delete req;        /* completed: slot already recycled */
+        else
+            req->consume = 1;  /* proxy frees it at completion */
+        *reqp = MPIX_REQUEST_NULL;
+        return MPI_SUCCESS;
+    }
 
     if (req->kind == ReqKind::BASIC) {
         int idx = req->flag_idx;

@@ -187,6 +187,33 @@ extern "C" int MPIX_Init(void)
     s->flags = reinterpret_cast<std::atomic<uint32_t> *>(raw);
     for (size_t i = 0; i < s->nflags; i++)
         s->flags[i].store(MPIX_FLAG_AVAILABLE, std::memory_order_relaxed);
+
+    /* fast-wait: second pinned word array for completion sequence numbers */
+    s->fast_wait = env_int("MPIX_FAST_WAIT", 0) != 0;
+    {
+        void *sraw = nullptr;
+        if (s->have_gpu) {
+            if (hipHostMalloc(&sraw, s->nflags * sizeof(uint32_t),
+                              hipHostMallocMapped) != hipSuccess) {
+                MPIX_ERR("hipHostMalloc(seq pool) failed");
+                delete s;
+                return MPI_ERR_OTHER;
+            }
+            void *sd = nullptr;
+            if (hipHostGetDevicePointer(&sd, sraw, 0) != hipSuccess) {
+                MPIX_ERR("hipHostGetDevicePointer(seq) failed");
+                delete s;
+                return MPI_ERR_OTHER;
+            }
+            s->seqs_d = (uint32_t *)sd;
+        } else {
+            sraw = calloc(s->nflags, sizeof(uint32_t));
+        }
+        s->seqs = reinterpret_cast<std::atomic<uint32_t> *>(sraw);
+        for (size_t i = 0; i < s->nflags; i++)
+            s->seqs[i].store(0, std::memory_order_relaxed);
+        s->slot_seq = (uint32_t *)calloc(s->nflags, sizeof(uint32_t));
+    }
     s->ops = new Op[s->nflags];
     s->armed.init(4 * s->nflags);
     s->spin_before_yield = env_int("MPIX_PROXY_SPIN", 2000);
@@ -276,6 +303,9 @@ extern "C" int MPIX_Finalize(void)
     delete[] s->ops;
     if (s->flags_pinned) (void)hipHostFree((void *)s->flags);
     else free((void *)s->flags);
+    if (s->flags_pinned) (void)hipHostFree((void *)s->seqs);
+    else free((void *)s->seqs);
+    free(s->slot_seq);
 
     g_state = nullptr;
     delete s;

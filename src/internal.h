@@ -107,6 +107,8 @@ struct Op {
     void *ch_priv = nullptr;        /* channel-private per-op state */
 
     /* completion/status delivery (guarded by g_state->completion_mutex) */
+    bool fast = false;              /* fast-wait protocol op */
+    bool waiter_owns_req = false;   /* fast: a host waiter will free req */
     MPI_Status *enq_status_target = nullptr; /* posted by MPIX_Wait*_enqueue */
     MPI_Status saved_status;
     bool status_saved = false;
@@ -123,6 +125,7 @@ struct Op {
         partition = -1; pseq = 0; req = nullptr;
         ch_done.store(0, std::memory_order_relaxed);
         ch_status = ChStatus{}; ch_priv = nullptr;
+        fast = false; waiter_owns_req = false;
         enq_status_target = nullptr; status_saved = false;
         orphaned.store(false, std::memory_order_relaxed);
     }
@@ -135,6 +138,16 @@ struct Request {
 
     /* BASIC */
     int flag_idx = -1;
+    /* fast-wait protocol (MPIX_FAST_WAIT=1): completion is a monotonic
+     * per-slot sequence number in a second pinned word array, waited with
+     * GTE — no CLEANUP write, slot freed by the proxy at completion.
+     * The proxy stores the status here (the op is recycled immediately). */
+    bool fast = false;
+    uint32_t seq = 0;
+    MPI_Status fast_status{};
+    /* who frees this request (guarded by completion_mutex):
+     * 0 = undecided, 1 = proxy at completion, 2 = a host waiter */
+    int consume = 0;
 
     /* PARTITIONED */
     bool is_send = false;
@@ -226,6 +239,10 @@ struct State {
     size_t nflags = 0;
     std::atomic<uint32_t> *flags = nullptr;  /* host view (pinned if GPU) */
     uint32_t *flags_d = nullptr;             /* device view or nullptr */
+    bool fast_wait = false;                  /* MPIX_FAST_WAIT=1 */
+    std::atomic<uint32_t> *seqs = nullptr;   /* fast: completion seq words */
+    uint32_t *seqs_d = nullptr;
+    uint32_t *slot_seq = nullptr;            /* fast: per-slot next seq */
     bool flags_pinned = false;
     Op *ops = nullptr;                       /* parallel op table */
     std::atomic<uint32_t> alloc_cursor{0};
@@ -258,6 +275,12 @@ static inline uint32_t flag_load(int idx) {
 }
 static inline void flag_store(int idx, uint32_t v) {
     g_state->flags[idx].store(v, std::memory_order_release);
+}
+static inline uint32_t seq_load(int idx) {
+    return g_state->seqs[idx].load(std::memory_order_acquire);
+}
+static inline void seq_store(int idx, uint32_t v) {
+    g_state->seqs[idx].store(v, std::memory_order_release);
 }
 static inline bool flag_cas(int idx, uint32_t expect, uint32_t v) {
     return g_state->flags[idx].compare_exchange_strong(

@@ -155,6 +155,27 @@ void proxy_main()
                 break;
             }
             case MPIX_FLAG_ISSUED:
+                if (op->fast &&
+                    op->ch_done.load(std::memory_order_acquire)) {
+                    /* fast-wait completion: status into the request, seq
+                     * word release-store (waiters use GTE), slot freed NOW
+                     * — no COMPLETED/CLEANUP round trip */
+                    Request *req = op->req;
+                    std::lock_guard<std::mutex> lk(s->completion_mutex);
+                    fill_status(&req->fast_status, op->ch_status);
+                    if (op->enq_status_target != nullptr)
+                        *op->enq_status_target = req->fast_status;
+                    bool free_req = (req->consume == 1);
+                    MPIX_TRACE_EV("slot %d fast complete seq=%u err=%d", idx,
+                                  req->seq, op->ch_status.err);
+                    seq_store(idx, req->seq);
+                    if (free_req) delete req;
+                    s->ops_completed.fetch_add(1, std::memory_order_relaxed);
+                    slot_free(idx);
+                    drop(i);
+                    did = true;
+                    continue;
+                }
                 if (op->ch_done.load(std::memory_order_acquire)) {
                     if (s->stats && op->t_issue_ns) {
                         uint64_t d = now_ns() - op->t_issue_ns;
