@@ -320,6 +320,13 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
     if (req->fast && qtype == MPIX_QUEUE_HIP_STREAM) {
         if (!s->have_gpu) return MPIX_Wait(reqp, status);
         hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
+        if (stream_capturing(stream)) {
+            /* the epoch wait value is request-specific; a captured graph
+             * would replay it stale (same reason graphs stay classic) */
+            MPIX_ERR("MPIX_FAST_WAIT requests cannot be waited inside a "
+                     "stream capture");
+            return MPI_ERR_REQUEST;
+        }
         {
             std::lock_guard<std::mutex> lk(s->completion_mutex);
             if (seq_load(idx) >= req->seq) {
@@ -350,6 +357,11 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
 
     if (qtype == MPIX_QUEUE_HIP_GRAPH) {
         if (!s->have_gpu) return MPI_ERR_OTHER;
+        if (req->fast) {
+            MPIX_ERR("MPIX_FAST_WAIT requests cannot be waited on a graph "
+                     "queue (graphs use the classic EQ protocol)");
+            return MPI_ERR_REQUEST;
+        }
         if (status != nullptr && status != MPI_STATUS_IGNORE) {
             std::lock_guard<std::mutex> lk(s->completion_mutex);
             Op *op = &s->ops[idx];
