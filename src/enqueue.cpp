@@ -440,6 +440,12 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         for (int i = 0; i < count; i++) {
             Request *req = (Request *)reqs[i];
             if (req == nullptr || req->kind != ReqKind::BASIC) continue;
+            if (req->fast) {
+                (void)hipGraphDestroy(g);
+                MPIX_ERR("MPIX_FAST_WAIT requests cannot be waited on a "
+                         "graph queue");
+                return MPI_ERR_REQUEST;
+            }
             int idx = req->flag_idx;
             {
                 std::lock_guard<std::mutex> lk(s->completion_mutex);
@@ -548,6 +554,11 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
             Request *req = (Request *)reqs[i];
             if (req == nullptr) continue;
             if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            if (req->fast) {
+                MPIX_ERR("MPIX_FAST_WAIT requests cannot be waited inside "
+                         "a stream capture");
+                return MPI_ERR_REQUEST;
+            }
             int idx = req->flag_idx;
             {
                 std::lock_guard<std::mutex> lk(s->completion_mutex);
