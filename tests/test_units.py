@@ -220,3 +220,40 @@ def test_query_config_shape(mpix_env):
                         "mpi_mode", "nflags"}
     assert cfg["nflags"] >= 64
     assert cfg["mpi_mode"] is False
+
+
+def test_nflags_env_respected(monkeypatch):
+    monkeypatch.setenv("RANK", "0")
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    monkeypatch.setenv("MPIX_NFLAGS", "256")
+    import mpix
+    mpix.init()
+    try:
+        assert mpix.config()["nflags"] == 256
+    finally:
+        mpix.finalize()
+
+
+def test_trace_and_stats_emit(monkeypatch):
+    """MPIX_TRACE / MPIX_STATS produce their diagnostics (run in a child so
+    the env-gated statics are evaluated fresh)."""
+    import subprocess, sys, os
+    code = (
+        "import numpy as np, mpix\n"
+        "mpix.init()\n"
+        "a=np.arange(8,dtype=np.int32); b=np.zeros(8,dtype=np.int32)\n"
+        "rs=mpix.isend_enqueue(a,dest=0,tag=1)\n"
+        "rr=mpix.irecv_enqueue(b,source=0,tag=1)\n"
+        "mpix.wait(rr); mpix.wait(rs)\n"
+        "mpix.finalize()\n"
+    )
+    env = {**os.environ, "RANK": "0", "WORLD_SIZE": "1",
+           "MPIX_TRACE": "1", "MPIX_STATS": "1",
+           "PYTHONPATH": os.path.dirname(os.path.dirname(
+               os.path.abspath(__file__)))}
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120, env=env)
+    assert r.returncode == 0, r.stderr
+    assert "[mpix trace" in r.stderr
+    assert "[mpix stats" in r.stderr
+    assert "latency histogram" in r.stderr
