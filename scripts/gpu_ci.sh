@@ -30,7 +30,25 @@ step gemm_check 90 bench/bin/gemm_pready --check
 step gemm_overlap 120 mpiexec -np 2 bench/bin/gemm_pready 4096 4096 4096 5
 
 if [ "$MODE" = full ]; then
+    # fast-wait A/B (epoch/GTE wait protocol) — round-2 measurement
+    note fastwait_pytest
+    MPIX_FAST_WAIT=1 timeout 180 python -m pytest tests/test_gpu.py -q -m gpu \
+        -k "loopback or partitioned" > gpurun_out/fastwait_pytest.log 2>&1 \
+        && echo OK || { echo "FAIL rc=$?"; tail -4 gpurun_out/fastwait_pytest.log; }
+    note fastwait_pingpong
+    MPIX_FAST_WAIT=1 timeout 100 mpiexec -np 2 bench/bin/pingpong 18 100 \
+        > gpurun_out/fastwait_pingpong.log 2>&1 \
+        && { echo OK; head -6 gpurun_out/fastwait_pingpong.log; } \
+        || echo "FAIL rc=$?"
+    note devpush_pingpong
+    MPIX_DEV_PUSH_MAX=65536 timeout 100 mpiexec -np 2 bench/bin/pingpong 15 50 \
+        > gpurun_out/devpush_pingpong.log 2>&1 \
+        && { echo OK; head -5 gpurun_out/devpush_pingpong.log; } \
+        || echo "FAIL rc=$?"
     step gemm_8k 90 bench/bin/gemm_pready 8192 8192 8192 5
+    step gemm_v3_check 90 bench/bin/gemm_pready_v3 --check
+    step gemm_v3_8k 90 bench/bin/gemm_pready_v3 8192 8192 8192 5
+    tail -1 gpurun_out/gemm_v3_8k.log
     note rocprof
     ( cd /tmp && timeout 150 rocprofv3 --kernel-trace --stats \
         -d "$OLDPWD/gpurun_out/prof" -o ci \
