@@ -89,19 +89,37 @@ static bool stream_capturing(hipStream_t stream)
 }
 
 /* hipUserObject destructor: releases an enqueued request owned by a graph.
- * Runs when the owning graph is destroyed.  Spins out any in-flight state,
- * then hands the slot to the proxy via CLEANUP. */
+ * HIP schedules this on an internal thread when the owning graph's refcount
+ * drops — possibly long after the graph was destroyed, after MPIX_Finalize,
+ * or even after a subsequent MPIX_Init built a NEW flag pool where this
+ * request's slot index means something else.  So: (a) never spin here (a
+ * wedged HIP callback thread stalls all later graph/user-object work in the
+ * process), and (b) act on the pool only if the state generation still
+ * matches.  In-flight slots are orphaned (the proxy frees them at
+ * completion); idle slots are handed over via CLEANUP. */
 static void graph_request_destroy(void *ud)
 {
     Request *req = (Request *)ud;
+    std::lock_guard<std::mutex> lg(lifecycle_mutex());
     State *s = g_state;
-    if (s == nullptr) { delete req; return; }
+    if (s == nullptr || s->gen != req->state_gen) {
+        delete req; /* the owning state (and its slots) are gone */
+        return;
+    }
     int idx = req->flag_idx;
-    uint32_t f;
-    while ((f = flag_load(idx)) == MPIX_FLAG_PENDING || f == MPIX_FLAG_ISSUED)
-        std::this_thread::yield();
-    /* RESERVED (never launched) or COMPLETED: proxy frees slot + request */
-    flag_store(idx, MPIX_FLAG_CLEANUP);
+    std::lock_guard<std::mutex> lk(s->completion_mutex);
+    if (idx < 0 || idx >= (int)s->nflags || s->ops[idx].req != req) {
+        delete req; /* slot already recycled past this request */
+        return;
+    }
+    uint32_t f = flag_load(idx);
+    if (f == MPIX_FLAG_PENDING || f == MPIX_FLAG_ISSUED) {
+        /* in flight: proxy deletes req + frees the slot at completion */
+        s->ops[idx].orphaned.store(true, std::memory_order_relaxed);
+    } else {
+        /* RESERVED (never launched) or COMPLETED: proxy frees slot + req */
+        flag_store(idx, MPIX_FLAG_CLEANUP);
+    }
 }
 
 /* Attach graph-owned cleanup for `req` to `graph`. */
@@ -234,6 +252,7 @@ static int enqueue_sendrecv(bool is_send, void *buf, int count,
 
     Request *req = new Request();
     req->kind = ReqKind::BASIC;
+    req->state_gen = s->gen;
     req->flag_idx = idx;
     op->req = req;
 
@@ -327,6 +346,7 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
                      "stream capture");
             return MPI_ERR_REQUEST;
         }
+        uint32_t wait_seq;
         {
             std::lock_guard<std::mutex> lk(s->completion_mutex);
             if (seq_load(idx) >= req->seq) {
@@ -337,18 +357,22 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
                 *reqp = MPIX_REQUEST_NULL;
                 return MPI_SUCCESS;
             }
+            /* copy everything we need BEFORE publishing consume=1: once the
+             * proxy sees consume==1 it deletes req at completion, which can
+             * happen the moment the mutex is released */
+            wait_seq = req->seq;
             req->consume = 1; /* proxy frees req at completion */
             if (status != nullptr && status != MPI_STATUS_IGNORE)
                 s->ops[idx].enq_status_target = status;
         }
         uint32_t *seq_d = s->seqs_d + idx;
         if (s->use_memops) {
-            MPIX_CHECK_HIP(hipStreamWaitValue32(stream, seq_d, req->seq,
+            MPIX_CHECK_HIP(hipStreamWaitValue32(stream, seq_d, wait_seq,
                                                 hipStreamWaitValueGte,
                                                 0xFFFFFFFFu));
         } else {
             hipLaunchKernelGGL(k_wait_flag_gte, dim3(1), dim3(1), 0, stream,
-                               seq_d, req->seq);
+                               seq_d, wait_seq);
             MPIX_CHECK_HIP(hipGetLastError());
         }
         *reqp = MPIX_REQUEST_NULL;
@@ -490,6 +514,17 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
             Request *req = (Request *)reqs[i];
             if (req == nullptr) continue;
             if (req->kind != ReqKind::BASIC) return MPI_ERR_REQUEST;
+            if (req->fast) {
+                /* fast ops complete via the seq word and free their slot
+                 * immediately — a batched EQ-COMPLETED wait on the recycled
+                 * flag would hang or fire on an unrelated op.  Route through
+                 * the GTE wait, same as the unbatched path. */
+                MPIX_Request r1 = (MPIX_Request)req;
+                MPIX_CHECK(MPIX_Wait_enqueue(&r1, status_at(i),
+                                             MPIX_QUEUE_HIP_STREAM, queue));
+                reqs[i] = MPIX_REQUEST_NULL;
+                continue;
+            }
             int idx = req->flag_idx;
             if (try_complete_now(idx, status_at(i))) {
                 reqs[i] = MPIX_REQUEST_NULL;
@@ -620,6 +655,17 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         if (req->kind != ReqKind::BASIC) {
             (void)hipHostFree(idx_arr);
             return MPI_ERR_REQUEST;
+        }
+        if (req->fast) { /* same rerouting as the other Waitall branches */
+            MPIX_Request r1 = (MPIX_Request)req;
+            int rc = MPIX_Wait_enqueue(&r1, status_at(i),
+                                       MPIX_QUEUE_HIP_STREAM, queue);
+            if (rc != MPI_SUCCESS) {
+                (void)hipHostFree(idx_arr);
+                return rc;
+            }
+            reqs[i] = MPIX_REQUEST_NULL;
+            continue;
         }
         int idx = req->flag_idx;
         if (try_complete_now(idx, status_at(i))) {

@@ -108,6 +108,27 @@ static void probe_memops(State *s)
     }
 }
 
+/* Free everything MPIX_Init may have allocated up to a failure point. */
+static void destroy_pools(State *s)
+{
+    delete[] s->ops;
+    s->ops = nullptr;
+    if (s->flags) {
+        if (s->flags_pinned) (void)hipHostFree((void *)s->flags);
+        else free((void *)s->flags);
+        s->flags = nullptr;
+    }
+    if (s->seqs) {
+        if (s->seqs_pinned) (void)hipHostFree((void *)s->seqs);
+        else free((void *)s->seqs);
+        s->seqs = nullptr;
+    }
+    free(s->slot_seq);
+    s->slot_seq = nullptr;
+}
+
+static std::atomic<uint64_t> g_init_gen{0};
+
 extern "C" int MPIX_Init(void)
 {
     if (g_state != nullptr) {
@@ -115,6 +136,7 @@ extern "C" int MPIX_Init(void)
         return MPI_ERR_OTHER;
     }
     State *s = new State();
+    s->gen = g_init_gen.fetch_add(1, std::memory_order_relaxed) + 1;
 
     /* --- identity: MPI mode or env mode ------------------------------- */
     int mpi_inited = 0, mpi_finalized = 0;
@@ -173,9 +195,11 @@ extern "C" int MPIX_Init(void)
             return MPI_ERR_OTHER;
         }
         s->flags_pinned = true;
+        s->flags = reinterpret_cast<std::atomic<uint32_t> *>(raw);
         void *dptr = nullptr;
         if (hipHostGetDevicePointer(&dptr, raw, 0) != hipSuccess) {
             MPIX_ERR("hipHostGetDevicePointer failed");
+            destroy_pools(s);
             delete s;
             return MPI_ERR_OTHER;
         }
@@ -196,12 +220,16 @@ extern "C" int MPIX_Init(void)
             if (hipHostMalloc(&sraw, s->nflags * sizeof(uint32_t),
                               hipHostMallocMapped) != hipSuccess) {
                 MPIX_ERR("hipHostMalloc(seq pool) failed");
+                destroy_pools(s);
                 delete s;
                 return MPI_ERR_OTHER;
             }
+            s->seqs_pinned = true;
+            s->seqs = reinterpret_cast<std::atomic<uint32_t> *>(sraw);
             void *sd = nullptr;
             if (hipHostGetDevicePointer(&sd, sraw, 0) != hipSuccess) {
                 MPIX_ERR("hipHostGetDevicePointer(seq) failed");
+                destroy_pools(s);
                 delete s;
                 return MPI_ERR_OTHER;
             }
@@ -219,7 +247,10 @@ extern "C" int MPIX_Init(void)
     s->spin_before_yield = env_int("MPIX_PROXY_SPIN", 2000);
     s->stats = env_int("MPIX_STATS", 0) != 0;
 
-    g_state = s; /* utilities below use g_state */
+    {
+        std::lock_guard<std::mutex> lg(lifecycle_mutex());
+        g_state = s; /* utilities below use g_state */
+    }
 
     /* --- memOps fast path ---------------------------------------------- */
     probe_memops(s);
@@ -229,7 +260,9 @@ extern "C" int MPIX_Init(void)
                                         s->mpi_mode, s->have_gpu, s->device_id);
     if (s->t_native == nullptr) {
         MPIX_ERR("native transport bring-up failed");
+        std::lock_guard<std::mutex> lg(lifecycle_mutex());
         g_state = nullptr;
+        destroy_pools(s);
         delete s;
         return MPI_ERR_OTHER;
     }
@@ -268,9 +301,24 @@ extern "C" int MPIX_Finalize(void)
     s->proxy_stop.store(true);
     if (s->proxy.joinable()) s->proxy.join();
 
+    /* From here on, state teardown must exclude late hipUserObject
+     * destructors (graph_request_destroy): they check g_state + generation
+     * under the same mutex. */
+    std::lock_guard<std::mutex> lg(lifecycle_mutex());
+
     size_t leaked = 0;
     for (size_t i = 0; i < s->nflags; i++) {
         uint32_t f = s->flags[i].load(std::memory_order_relaxed);
+        if (f == MPIX_FLAG_CLEANUP) {
+            /* consumed after the proxy exited (e.g. a graph destroyed right
+             * before finalize): free the request here */
+            Op *op = &s->ops[i];
+            if ((op->kind == OpKind::ISEND || op->kind == OpKind::IRECV) &&
+                op->req != nullptr)
+                delete op->req;
+            s->flags[i].store(MPIX_FLAG_AVAILABLE, std::memory_order_relaxed);
+            continue;
+        }
         if (f != MPIX_FLAG_AVAILABLE) leaked++;
     }
     if (leaked)
@@ -300,12 +348,7 @@ extern "C" int MPIX_Finalize(void)
     }
     delete s->t_mpi;
 
-    delete[] s->ops;
-    if (s->flags_pinned) (void)hipHostFree((void *)s->flags);
-    else free((void *)s->flags);
-    if (s->flags_pinned) (void)hipHostFree((void *)s->seqs);
-    else free((void *)s->seqs);
-    free(s->slot_seq);
+    destroy_pools(s);
 
     g_state = nullptr;
     delete s;

@@ -135,6 +135,12 @@ enum class ReqKind : int32_t { BASIC = 0, PARTITIONED };
 
 struct Request {
     ReqKind kind = ReqKind::BASIC;
+    /* init-generation of the State this request belongs to.  hipUserObject
+     * destructors for graph-owned requests can fire on a HIP-internal thread
+     * at any later time — including after MPIX_Finalize or a following
+     * MPIX_Init — so they must verify the state generation before touching
+     * the flag pool (see graph_request_destroy). */
+    uint64_t state_gen = 0;
 
     /* BASIC */
     int flag_idx = -1;
@@ -229,6 +235,7 @@ struct State {
     /* identity */
     int world_rank = 0;
     int world_size = 1;
+    uint64_t gen = 0;           /* init-generation (see lifecycle_mutex) */
     bool mpi_mode = false;      /* MPI_Init was called by the app */
     /* gpu */
     bool have_gpu = false;
@@ -244,6 +251,7 @@ struct State {
     uint32_t *seqs_d = nullptr;
     uint32_t *slot_seq = nullptr;            /* fast: per-slot next seq */
     bool flags_pinned = false;
+    bool seqs_pinned = false;
     Op *ops = nullptr;                       /* parallel op table */
     std::atomic<uint32_t> alloc_cursor{0};
     /* proxy */
@@ -267,6 +275,12 @@ struct State {
 };
 
 extern State *g_state;
+
+/* Serializes MPIX_Init/Finalize state publication against asynchronously
+ * scheduled hipUserObject destructors (graph-owned request cleanup).  A
+ * leaked function-local static so it outlives any HIP runtime thread that
+ * might still fire a destructor during process exit. */
+std::mutex &lifecycle_mutex();
 
 /* ------------------------------------------------------------- flag helpers */
 

@@ -88,15 +88,28 @@ void proxy_main()
     std::vector<int> watch;
     watch.reserve(s->nflags);
     std::vector<uint8_t> watched(s->nflags, 0);
+    size_t dead = 0; /* tombstones in `watch` */
     int idle = 0;
 
     /* Order-preserving removal: the walk below must see slots in ARM order,
      * because two sends to one peer that both read PENDING in the same pass
      * are issued in walk order — swap-remove compaction reordered the list
-     * and broke per-(src,tag) FIFO (caught by tests/test_soak.py). */
+     * and broke per-(src,tag) FIFO (caught by tests/test_soak.py).
+     * vector::erase kept order but cost O(n) per removal (quadratic under
+     * churn with many active slots), so removal is a tombstone and the list
+     * is compacted — still in order — once mostly dead. */
     auto drop = [&](size_t i) {
         watched[watch[i]] = 0;
-        watch.erase(watch.begin() + (long)i);
+        watch[i] = -1;
+        dead++;
+    };
+    auto compact = [&] {
+        if (dead < 64 || dead * 2 < watch.size()) return;
+        size_t w = 0;
+        for (size_t i = 0; i < watch.size(); i++)
+            if (watch[i] >= 0) watch[w++] = watch[i];
+        watch.resize(w);
+        dead = 0;
     };
 
     while (true) {
@@ -117,8 +130,9 @@ void proxy_main()
         if (s->t_mpi) s->t_mpi->progress();
 
         /* walk active slots */
-        for (size_t i = 0; i < watch.size();) {
+        for (size_t i = 0; i < watch.size(); i++) {
             int idx = watch[i];
+            if (idx < 0) continue; /* tombstone */
             Op *op = &s->ops[idx];
             uint32_t f = flag_load(idx);
             switch (f) {
@@ -227,10 +241,11 @@ void proxy_main()
             default:
                 break;
             }
-            i++;
         }
+        compact();
 
-        if (s->proxy_stop.load(std::memory_order_acquire) && watch.empty())
+        if (s->proxy_stop.load(std::memory_order_acquire) &&
+            watch.size() == dead)
             break;
 
         if (did) {
@@ -246,6 +261,7 @@ void proxy_main()
                      * app is exiting without waiting (leak warning follows) */
                     bool in_flight = false;
                     for (int idx2 : watch) {
+                        if (idx2 < 0) continue; /* tombstone */
                         uint32_t f2 = flag_load(idx2);
                         if (f2 == MPIX_FLAG_PENDING || f2 == MPIX_FLAG_ISSUED ||
                             f2 == MPIX_FLAG_CLEANUP) { in_flight = true; break; }

@@ -13,6 +13,14 @@ namespace mpix {
 
 State *g_state = nullptr;
 
+std::mutex &lifecycle_mutex()
+{
+    /* intentionally leaked: HIP user-object destructors may fire after
+     * static destructors have begun */
+    static std::mutex *m = new std::mutex();
+    return *m;
+}
+
 /* ------------------------------------------------------------------- slots */
 
 int slot_allocate()

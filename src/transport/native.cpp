@@ -748,7 +748,14 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
     }
     uint64_t n = st.bytes;
     hipError_t e = hipSuccess;
-    if (n > 0 && n <= copy_kernel_max()) {
+    /* The shader pull path requires (a) a device destination — dereferencing
+     * a pageable host pointer from a kernel faults on non-XNACK systems —
+     * and (b) 16-byte alignment on both sides (user buffers can be
+     * arbitrarily offset, e.g. partitioned slices).  Everything else rides
+     * hipMemcpyAsync (SDMA / runtime blit). */
+    bool kernel_ok = op->buf_is_device &&
+                     ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
+    if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
         unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
         if (blocks == 0) blocks = 1;

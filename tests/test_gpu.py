@@ -115,6 +115,10 @@ def test_graph_construction_loopback(mpix_env):
         assert (recv == 100 + it).all(), f"iter {it}"
     mpix.graph_exec_destroy(gexec)
     mpix.graph_destroy(parent)
+    # child graphs were cloned into the parent; drop our references so the
+    # request user-objects can fire their destructors
+    for g in (g_send, g_recv, g_wait):
+        mpix.graph_destroy(g)
 
 
 def test_partitioned_device_kernels(mpix_env):
@@ -172,7 +176,7 @@ def _ring_device_stream(rank, size):
 
 
 def test_ring_device_2proc():
-    run_ranks(2, _ring_device_stream, timeout=300)
+    run_ranks(2, _ring_device_stream, timeout=240)
 
 
 def _pingpong_device(rank, size):
@@ -207,7 +211,7 @@ def _pingpong_device(rank, size):
 
 
 def test_pingpong_device_2proc():
-    run_ranks(2, _pingpong_device, timeout=300)
+    run_ranks(2, _pingpong_device, timeout=240)
 
 
 def _partitioned_ring_device(rank, size):
@@ -248,4 +252,4 @@ def _partitioned_ring_device(rank, size):
 
 
 def test_partitioned_ring_device_2proc():
-    run_ranks(2, _partitioned_ring_device, timeout=300)
+    run_ranks(2, _partitioned_ring_device, timeout=240)
