@@ -339,6 +339,38 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
 
 #endif /* MPIX_GEMM_VARIANT */
 
+/* ---------------------------------------------------- payload verification
+ * Sampled end-to-end check of the RECEIVED buffer in timed mode (not just
+ * --check of the GEMM): every VERIFY_EVERY iterations the receiver's Crecv
+ * partitions are poisoned before Start, then after the waits an xor
+ * checksum of Crecv is compared against the checksum of the left rank's C
+ * (exchanged over MPI).  A stale/skipped transfer leaves the poison words
+ * in place and fails the compare; corrupt data fails it outright. */
+#define VERIFY_EVERY 4
+
+__global__ void k_xorsum(const uint32_t *p, size_t n_words,
+                         unsigned long long *out)
+{
+    __shared__ unsigned long long blk;
+    if (threadIdx.x == 0) blk = 0;
+    __syncthreads();
+    unsigned long long acc = 0;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n_words; i += stride)
+        /* fold the word index in so permuted data doesn't cancel out */
+        acc ^= (unsigned long long)p[i] * 2654435761ull + i;
+    atomicXor(&blk, acc);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicXor(out, blk);
+}
+
+__global__ void k_poison(uint32_t *buf, int nparts, size_t part_words)
+{
+    int p = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+    if (p < nparts) buf[(size_t)p * part_words] = 0xDEADBEEFu;
+}
+
 static void host_gemm_ref(const std::vector<float> &A,
                           const std::vector<float> &B, std::vector<float> &C,
                           int M, int N, int K)
@@ -448,8 +480,19 @@ int main(int argc, char **argv)
     MPIX_Prequest dps;
     CHECK(MPIX_Prequest_create(ps, &dps) == 0);
 
+    unsigned long long *d_sum; /* [0]=xorsum(C), [1]=xorsum(Crecv) */
+    HIP(hipMalloc(&d_sum, 2 * sizeof(unsigned long long)));
+    size_t sum_words = cn * sizeof(bf16) / 4;
+    size_t part_words = sum_words / nparts;
+
     auto run = [&](bool overlap) {
-        auto one = [&]() {
+        auto one = [&](int it) {
+            bool verify = (it % VERIFY_EVERY) == 0;
+            if (verify) {
+                hipLaunchKernelGGL(k_poison, dim3(1), dim3(64), 0, st,
+                                   (uint32_t *)Crecv, nparts, part_words);
+                HIP(hipStreamSynchronize(st)); /* land before the pull */
+            }
             MPIX_Request act[2] = {pr, ps};
             CHECK(MPIX_Startall(2, act) == 0);
             HIP(hipMemsetAsync(band_cnt, 0, nparts * sizeof(uint32_t), st));
@@ -468,12 +511,32 @@ int main(int argc, char **argv)
             }
             CHECK(MPIX_Wait(&pr, MPI_STATUS_IGNORE) == 0);
             CHECK(MPIX_Wait(&ps, MPI_STATUS_IGNORE) == 0);
+            if (verify) {
+                HIP(hipMemsetAsync(d_sum, 0, 2 * sizeof(unsigned long long),
+                                   st));
+                hipLaunchKernelGGL(k_xorsum, dim3(256), dim3(256), 0, st,
+                                   (const uint32_t *)C, sum_words, d_sum);
+                hipLaunchKernelGGL(k_xorsum, dim3(256), dim3(256), 0, st,
+                                   (const uint32_t *)Crecv, sum_words,
+                                   d_sum + 1);
+                unsigned long long h[2], left_sum = 0;
+                HIP(hipMemcpy(h, d_sum, sizeof(h), hipMemcpyDeviceToHost));
+                MPI_Sendrecv(&h[0], 1, MPI_UNSIGNED_LONG_LONG, right, 77,
+                             &left_sum, 1, MPI_UNSIGNED_LONG_LONG, left, 77,
+                             MPI_COMM_WORLD, MPI_STATUS_IGNORE);
+                if (left_sum != h[1]) {
+                    fprintf(stderr, "[r%d] VERIFY FAIL iter %d: Crecv xorsum "
+                            "%016llx != sender C %016llx (overlap=%d)\n",
+                            rank, it, h[1], left_sum, (int)overlap);
+                    MPI_Abort(MPI_COMM_WORLD, 2);
+                }
+            }
         };
-        for (int i = 0; i < warmup; i++) one();
+        for (int i = 0; i < warmup; i++) one(i);
         HIP(hipDeviceSynchronize());
         MPI_Barrier(MPI_COMM_WORLD);
         auto t0 = std::chrono::steady_clock::now();
-        for (int i = 0; i < iters; i++) one();
+        for (int i = 0; i < iters; i++) one(i);
         MPI_Barrier(MPI_COMM_WORLD);
         double dt = std::chrono::duration<double>(
                         std::chrono::steady_clock::now() - t0).count();
@@ -511,7 +574,7 @@ int main(int argc, char **argv)
     CHECK(MPIX_Request_free(&ps) == 0);
     CHECK(MPIX_Request_free(&pr) == 0);
     (void)hipFree(A); (void)hipFree(B); (void)hipFree(C);
-    (void)hipFree(Crecv); (void)hipFree(band_cnt);
+    (void)hipFree(Crecv); (void)hipFree(band_cnt); (void)hipFree(d_sum);
     (void)hipStreamDestroy(st);
     MPIX_Finalize();
     MPI_Finalize();

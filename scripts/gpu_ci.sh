@@ -19,9 +19,16 @@ step() { # step <name> <timeout_s> <cmd...>
     fi
 }
 
-step pytest_gpu 300 python -m pytest tests -q -m gpu
-step bench_1gpu 150 python bench.py --steps 20 --warmup 5
-tail -1 gpurun_out/bench_1gpu.log
+step pytest_gpu 600 python -m pytest tests -q -m gpu
+step smoke 120 python -c 'from __graft_entry__ import smoke; smoke()'
+# headline stability: 5 back-to-back repetitions -> median +- spread
+note bench_repeat
+for i in 1 2 3 4 5; do
+    timeout 150 python bench.py --steps 20 --warmup 5 \
+        > "gpurun_out/bench_rep$i.log" 2>&1 \
+        && tail -1 "gpurun_out/bench_rep$i.log" || echo "rep$i FAIL rc=$?"
+done
+cp gpurun_out/bench_rep1.log gpurun_out/bench_1gpu.log 2>/dev/null || true
 
 export PATH=/opt/conda/bin:$PATH
 step pingpong 120 mpiexec -np 2 bench/bin/pingpong 24 100
