@@ -44,4 +44,14 @@ $(LIB): $(OBJS)
 clean:
 	rm -f $(OBJS) $(LIB) $(PYEXT)
 
-.PHONY: all clean python
+# Compile-validate the MPI-4.0 partitioned passthrough on this MPI-3.1
+# toolchain (prototypes declared by internal.h under the forced gate; the
+# objects are NOT linked into the library).  On a real MPI-4 library the
+# gate turns on automatically — see internal.h.
+mpi4check:
+	$(HIPCC) $(CXXFLAGS) -DMPIX_MPI_PARTITIONED -c src/partitioned.cpp -o /tmp/mpix_p4_part.o
+	$(HIPCC) $(CXXFLAGS) -DMPIX_MPI_PARTITIONED -c src/transport/mpi.cpp -o /tmp/mpix_p4_mpi.o
+	$(HIPCC) $(CXXFLAGS) -DMPIX_MPI_PARTITIONED -c src/enqueue.cpp -o /tmp/mpix_p4_enq.o
+	@echo "mpi4check: MPI-4 partitioned passthrough compiles both ways"
+
+.PHONY: all clean python mpi4check

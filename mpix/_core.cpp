@@ -19,6 +19,7 @@
 
 #include "mpix/mpix.h"
 #include "mpix/mpix_device.h"
+#include "../src/status_codec.h"
 
 namespace py = pybind11;
 
@@ -58,7 +59,7 @@ static py::dict status_to_dict(const MPI_Status &st)
     d["source"] = st.MPI_SOURCE;
     d["tag"] = st.MPI_TAG;
     d["error"] = st.MPI_ERROR;
-    d["count_bytes"] = st.count_lo; /* MPICH: byte count in count_lo */
+    d["count_bytes"] = mpix::status_bytes(st); /* ABI codec, status_codec.h */
     return d;
 }
 
@@ -177,6 +178,14 @@ PYBIND11_MODULE(_C, m)
         return py::make_tuple(
             py::cast(r, py::return_value_policy::take_ownership),
             (uintptr_t)g);
+    });
+    m.def("wait_graph", [](MxRequest *r) {
+        /* per-request wait graph (the reference's composition style in
+         * test/src/ring-all-graph-construction.c uses one per request) */
+        hipGraph_t g = nullptr;
+        PY_CHECK(MPIX_Wait_enqueue(&r->req, MPI_STATUS_IGNORE,
+                                   MPIX_QUEUE_HIP_GRAPH, &g));
+        return (uintptr_t)g;
     });
     m.def("waitall_graph", [](std::vector<MxRequest *> reqs) {
         std::vector<MPIX_Request> rr;
@@ -307,6 +316,12 @@ PYBIND11_MODULE(_C, m)
         hipGraphExec_t exec = nullptr;
         PY_CHECK_HIP(hipGraphInstantiate(&exec, parent, nullptr, nullptr, 0));
         return py::make_tuple((uintptr_t)parent, (uintptr_t)exec);
+    });
+    m.def("graph_instantiate", [](uintptr_t g) {
+        hipGraphExec_t exec = nullptr;
+        PY_CHECK_HIP(hipGraphInstantiate(&exec, (hipGraph_t)g, nullptr,
+                                         nullptr, 0));
+        return (uintptr_t)exec;
     });
     m.def("graph_launch", [](uintptr_t exec, uintptr_t stream) {
         PY_CHECK_HIP(hipGraphLaunch((hipGraphExec_t)exec,

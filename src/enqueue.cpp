@@ -756,6 +756,13 @@ static int host_wait_partitioned(Request *req, MPI_Status *status)
         flag_store(idx, MPIX_FLAG_RESERVED);
     }
     req->active = false;
+#if MPIX_HAVE_MPI_PARTITIONED
+    if (req->mpi_part_native) {
+        /* reference semantics (sendrecv.cu:627): the host wait completes the
+         * underlying persistent MPI request after all partitions are done */
+        MPIX_CHECK(MPI_Wait(&req->mpi_preq, MPI_STATUS_IGNORE));
+    }
+#endif
     if (status != nullptr && status != MPI_STATUS_IGNORE) {
         ChStatus cs;
         cs.src = req->is_send ? -1 : req->peer;
@@ -837,6 +844,10 @@ extern "C" int MPIX_Request_free(MPIX_Request *reqp)
     }
     if (req->dev_idx) (void)hipFree(req->dev_idx);
     if (req->dev_handle) (void)hipFree(req->dev_handle);
+#if MPIX_HAVE_MPI_PARTITIONED
+    if (req->mpi_part_native && req->mpi_preq != MPI_REQUEST_NULL)
+        (void)MPI_Request_free(&req->mpi_preq);
+#endif
     delete req;
     *reqp = MPIX_REQUEST_NULL;
     return MPI_SUCCESS;

@@ -31,6 +31,32 @@
 #include "mpix/mpix.h"
 #include "mpix/mpix_abi.h"
 
+/* --------------------------------------------- MPI-4.0 partitioned support
+ * Reference parity with /root/reference/Makefile:17-20: when the host MPI
+ * implements MPI-4.0 partitioned communication, MPIX_Psend/Precv_init can
+ * delegate to MPI_Psend_init/MPI_Pready instead of mpix's own emulation
+ * (opt-in at runtime with MPIX_MPI_PARTITIONED=1, MPI mode only).  The gate
+ * is automatic on an MPI-4 library; -DMPIX_MPI_PARTITIONED force-compiles
+ * the passthrough against declared prototypes on an MPI-3.x toolchain
+ * (compile validation only — calling it there aborts at link/bind time). */
+#if MPI_VERSION >= 4 || defined(MPIX_MPI_PARTITIONED)
+#define MPIX_HAVE_MPI_PARTITIONED 1
+#else
+#define MPIX_HAVE_MPI_PARTITIONED 0
+#endif
+#if MPIX_HAVE_MPI_PARTITIONED && MPI_VERSION < 4
+extern "C" {
+int MPI_Psend_init(const void *buf, int partitions, MPI_Count count,
+                   MPI_Datatype datatype, int dest, int tag, MPI_Comm comm,
+                   MPI_Info info, MPI_Request *request);
+int MPI_Precv_init(void *buf, int partitions, MPI_Count count,
+                   MPI_Datatype datatype, int source, int tag, MPI_Comm comm,
+                   MPI_Info info, MPI_Request *request);
+int MPI_Pready(int partition, MPI_Request request);
+int MPI_Parrived(MPI_Request request, int partition, int *flag);
+}
+#endif
+
 namespace mpix {
 
 /* ------------------------------------------------------------------ errors */
@@ -170,6 +196,12 @@ struct Request {
     uint32_t start_seq = 0;         /* number of MPIX_Start calls */
     mpix_prequest_dev_t *dev_handle = nullptr; /* device copy (Prequest_create) */
     int32_t *dev_idx = nullptr;                /* device idx array */
+    /* MPI-4.0 native partitioned passthrough (MPIX_HAVE_MPI_PARTITIONED +
+     * MPIX_MPI_PARTITIONED=1): the underlying persistent MPI request.  The
+     * per-partition flag protocol is unchanged; the proxy routes Pready /
+     * Parrived-polls to MPI instead of moving bytes itself. */
+    bool mpi_part_native = false;
+    MPI_Request mpi_preq = MPI_REQUEST_NULL;
 };
 
 /* ----------------------------------------------------------------- channel */

@@ -37,13 +37,27 @@ static int psend_precv_init(bool is_send, void *buf, int partitions,
     uint32_t comm_id = 0;
     bool native_ok = false;
     MPIX_CHECK(resolve_peer(comm, peer, &peer_world, &comm_id, &native_ok));
-    if (!native_ok) {
-        MPIX_ERR("partitioned ops require MPI_COMM_WORLD/SELF");
-        return MPI_ERR_COMM;
-    }
     if (peer_world == MPI_ANY_SOURCE) {
         MPIX_ERR("partitioned recv requires a concrete source rank");
         return MPI_ERR_RANK;
+    }
+
+    /* MPI-4.0 native passthrough (reference parity, partitioned.cu:57-59):
+     * runtime opt-in, MPI mode only.  The flag protocol stays ours; the MPI
+     * library owns transfer + matching, so arbitrary comms work too. */
+    bool mpi_native = false;
+#if MPIX_HAVE_MPI_PARTITIONED
+    {
+        static const int want = [] {
+            const char *e = getenv("MPIX_MPI_PARTITIONED");
+            return e ? atoi(e) : 0;
+        }();
+        mpi_native = want && s->mpi_mode;
+    }
+#endif
+    if (!native_ok && !s->mpi_mode) {
+        MPIX_ERR("partitioned ops on non-WORLD/SELF comms require MPI mode");
+        return MPI_ERR_COMM;
     }
 
     Request *req = new Request();
@@ -61,6 +75,24 @@ static int psend_precv_init(bool is_send, void *buf, int partitions,
     req->datatype = datatype;
     req->count_per_part = (int)count;
     req->active = false;
+
+#if MPIX_HAVE_MPI_PARTITIONED
+    if (mpi_native) {
+        int rc = is_send
+            ? MPI_Psend_init(buf, partitions, count, datatype, peer, tag,
+                             comm, MPI_INFO_NULL, &req->mpi_preq)
+            : MPI_Precv_init(buf, partitions, count, datatype, peer, tag,
+                             comm, MPI_INFO_NULL, &req->mpi_preq);
+        if (rc != MPI_SUCCESS) {
+            MPIX_ERR("MPI_P%s_init failed (%d)", is_send ? "send" : "recv", rc);
+            delete req;
+            return rc;
+        }
+        req->mpi_part_native = true;
+    }
+#else
+    (void)mpi_native;
+#endif
 
     for (int p = 0; p < partitions; p++) {
         int idx = slot_allocate();
@@ -84,7 +116,9 @@ static int psend_precv_init(bool is_send, void *buf, int partitions,
         op->comm = comm;
         op->comm_id = comm_id;
         op->partition = p;
-        op->native_route = true;
+        /* native shm/xGMI for WORLD/SELF; MPI transport for arbitrary comms
+         * (header-routed partition messages) and the MPI-4.0 passthrough */
+        op->native_route = native_ok && !req->mpi_part_native;
         op->req = req;
     }
     /* buffer kind determined once (partitions are slices of one buffer) */
@@ -129,6 +163,16 @@ extern "C" int MPIX_Start(MPIX_Request *reqp)
     }
     req->active = true;
     req->start_seq++;
+#if MPIX_HAVE_MPI_PARTITIONED
+    if (req->mpi_part_native) {
+        int rc = MPI_Start(&req->mpi_preq);
+        if (rc != MPI_SUCCESS) {
+            req->active = false;
+            req->start_seq--;
+            return rc;
+        }
+    }
+#endif
     for (int p = 0; p < req->n_partitions; p++) {
         int idx = req->part_idx[p];
         Op *op = &s->ops[idx];

@@ -8,6 +8,7 @@
 #include <hip/hip_runtime.h>
 
 #include "internal.h"
+#include "status_codec.h"
 
 namespace mpix {
 
@@ -62,15 +63,11 @@ int datatype_size(MPI_Datatype dt, int *size_out)
         *size_out = sz;
         return MPI_SUCCESS;
     }
-    /* MPICH encodes the size of builtin datatypes in handle bits 8..15
-     * (e.g. MPI_INT = 0x4c000405 -> 4 bytes).  Valid without MPI_Init. */
-    uintptr_t h = (uintptr_t)dt;
-    if ((h & 0xff000000u) == 0x4c000000u) {
-        int sz = (int)((h >> 8) & 0xffu);
-        if (sz > 0) { *size_out = sz; return MPI_SUCCESS; }
-    }
+    /* env mode (no MPI_Init): ABI-specific decode, see status_codec.h */
+    int sz = builtin_datatype_size(dt);
+    if (sz > 0) { *size_out = sz; return MPI_SUCCESS; }
     MPIX_ERR("non-builtin datatype requires MPI mode (handle 0x%lx)",
-             (unsigned long)h);
+             (unsigned long)(uintptr_t)dt);
     return MPI_ERR_TYPE;
 }
 
@@ -119,14 +116,7 @@ int resolve_peer(MPI_Comm comm, int rank, int *world_rank_out,
 void fill_status(MPI_Status *st, const ChStatus &cs)
 {
     if (st == nullptr) return;
-    /* MPICH 3.x layout: count_lo carries the received byte count, so
-     * MPI_Get_count(st, dtype, ..) = count_lo / type_size works. */
-    memset(st, 0, sizeof(*st));
-    st->count_lo = (int)cs.bytes;
-    st->count_hi_and_cancelled = 0;
-    st->MPI_SOURCE = cs.src;
-    st->MPI_TAG = cs.tag;
-    st->MPI_ERROR = cs.err;
+    status_encode(st, cs.src, cs.tag, cs.err, cs.bytes);
 }
 
 bool ptr_is_device(const void *ptr)

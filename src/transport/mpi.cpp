@@ -2,26 +2,114 @@
  *
  * Used only in MPI mode, for communicators other than WORLD/SELF (the native
  * shm/xGMI channel has no registered matching context for them).  Mirrors the
- * reference's proxy-issued MPI_Isend/Irecv/Test (init.cpp:69-140), with one
- * addition: because the sandbox MPICH 3.3.2 is not GPU-aware, device buffers
- * are staged automatically through pinned host bounce buffers.
+ * reference's proxy-issued MPI_Isend/Irecv/Test (init.cpp:69-140), plus:
+ *
+ *  - device buffers are staged through pinned host bounce buffers (the
+ *    sandbox MPICH 3.3.2 is not GPU-aware);
+ *  - partitioned ops on arbitrary communicators (the reference requires
+ *    MPI-4.0 for these, partitioned.cu:57-59).  Two modes:
+ *      (a) MPI-4.0 native passthrough (MPIX_MPI_PARTITIONED=1 on an MPI-4
+ *          library): the proxy routes Pready/Parrived-poll to MPI_Pready /
+ *          MPI_Parrived on the persistent request;
+ *      (b) MPI-3.1 emulation: each partition travels as one header-prefixed
+ *          MPI message on a reserved tag (MPI_TAG_UB); the header carries
+ *          {user tag, partition, start-seq} so arrival order never matters.
+ *          Receives are probe-driven (no pre-posted pool), so partition
+ *          sizes may differ freely across concurrent requests.
  */
 #include <hip/hip_runtime.h>
 
 #include <list>
+#include <vector>
 
 #include "../internal.h"
 
 namespace mpix {
 
+/* partition-message header (emulation mode) */
+struct PartHdr {
+    uint32_t magic = 0x4d505850u; /* "MPXP" */
+    int32_t partition = -1;
+    int32_t tag = 0;
+    uint32_t pseq = 0;
+    uint64_t bytes = 0;
+};
+static_assert(sizeof(PartHdr) == 24, "PartHdr layout");
+
+static int reserved_part_tag()
+{
+    static int tag = [] {
+        void *v = nullptr;
+        int found = 0, t = 32767; /* MPI minimum guarantee as fallback */
+        if (MPI_Comm_get_attr(MPI_COMM_WORLD, MPI_TAG_UB, &v, &found) ==
+                MPI_SUCCESS && found && v != nullptr)
+            t = *(int *)v;
+        return t;
+    }();
+    return tag;
+}
+
 class MpiTransport : public Transport {
 public:
     int start(Op *op) override {
-        if (op->kind == OpKind::PSEND_PART || op->kind == OpKind::PRECV_PART) {
-            MPIX_ERR("partitioned ops require MPI_COMM_WORLD/SELF "
-                     "(native channel)");
+        switch (op->kind) {
+        case OpKind::ISEND:
+        case OpKind::IRECV:
+            return start_basic(op);
+        case OpKind::PSEND_PART:
+            return start_psend_part(op);
+        case OpKind::PRECV_PART:
+            return start_precv_part(op);
+        default:
             return -1;
         }
+    }
+
+    void progress() override {
+        progress_basic();
+        progress_parrived();
+        progress_part_probe();
+    }
+
+    const char *name() const override { return "mpi-passthrough"; }
+
+private:
+    struct Out {
+        Op *op = nullptr;
+        MPI_Request req = MPI_REQUEST_NULL;
+        void *bounce = nullptr;
+        bool bounce_pinned = false;
+        bool is_recv = false;
+        bool is_part_send = false;
+    };
+    static void free_bounce(Out &o) {
+        if (!o.bounce) return;
+        if (o.bounce_pinned) (void)hipHostFree(o.bounce);
+        else free(o.bounce);
+        o.bounce = nullptr;
+    }
+    std::list<Out> outstanding_;
+    std::vector<Op *> parrived_;      /* MPI-4 passthrough recv partitions */
+    std::vector<Op *> part_recvs_;    /* emulation: pending PRECV_PART ops */
+    struct PartMsg {                  /* emulation: early partition arrivals */
+        MPI_Comm comm;
+        int src;
+        PartHdr hdr;
+        std::vector<char> payload;
+    };
+    std::list<PartMsg> part_unexpected_;
+
+    static void complete(Op *op, int src, int tag, uint64_t bytes, int err) {
+        op->ch_status.src = src;
+        op->ch_status.tag = tag;
+        op->ch_status.bytes = bytes;
+        op->ch_status.err = err;
+        op->ch_done.store(1, std::memory_order_release);
+    }
+
+    /* ------------------------------------------------------ basic ops */
+
+    int start_basic(Op *op) {
         Out o;
         o.op = op;
         o.is_recv = (op->kind == OpKind::IRECV);
@@ -32,10 +120,11 @@ public:
                 MPIX_ERR("bounce alloc failed (%lu B)", (unsigned long)op->bytes);
                 return -1;
             }
+            o.bounce_pinned = true;
             if (!o.is_recv &&
                 hipMemcpy(o.bounce, op->buf, op->bytes, hipMemcpyDeviceToHost)
                     != hipSuccess) {
-                (void)hipHostFree(o.bounce);
+                free_bounce(o);
                 return -1;
             }
             buf = o.bounce;
@@ -48,21 +137,21 @@ public:
             rc = MPI_Isend(buf, op->count, op->datatype, op->peer, op->tag,
                            op->comm, &o.req);
         if (rc != MPI_SUCCESS) {
-            if (o.bounce) (void)hipHostFree(o.bounce);
+            free_bounce(o);
             return -1;
         }
         outstanding_.push_back(o);
         return 0;
     }
 
-    void progress() override {
+    void progress_basic() {
         for (auto it = outstanding_.begin(); it != outstanding_.end();) {
             int done = 0;
             MPI_Status st;
             if (MPI_Test(&it->req, &done, &st) != MPI_SUCCESS) {
                 it->op->ch_status.err = MPI_ERR_OTHER;
                 it->op->ch_done.store(1, std::memory_order_release);
-                if (it->bounce) (void)hipHostFree(it->bounce);
+                free_bounce(*it);
                 it = outstanding_.erase(it);
                 continue;
             }
@@ -71,7 +160,9 @@ public:
                 continue;
             }
             Op *op = it->op;
-            if (it->is_recv) {
+            if (it->is_part_send) {
+                complete(op, -1, op->tag, op->bytes, MPI_SUCCESS);
+            } else if (it->is_recv) {
                 int cnt = 0, tsz = 0;
                 MPI_Get_count(&st, op->datatype, &cnt);
                 datatype_size(op->datatype, &tsz);
@@ -81,32 +172,189 @@ public:
                         (void)hipMemcpy(op->buf, it->bounce, n,
                                         hipMemcpyHostToDevice);
                 }
-                op->ch_status.src = st.MPI_SOURCE;
-                op->ch_status.tag = st.MPI_TAG;
-                op->ch_status.bytes = (uint64_t)cnt * (uint64_t)tsz;
-                op->ch_status.err = st.MPI_ERROR;
+                complete(op, st.MPI_SOURCE, st.MPI_TAG,
+                         (uint64_t)cnt * (uint64_t)tsz, st.MPI_ERROR);
             } else {
-                op->ch_status.src = -1; /* send status: fields undefined */
-                op->ch_status.tag = op->tag;
-                op->ch_status.bytes = op->bytes;
-                op->ch_status.err = MPI_SUCCESS;
+                complete(op, -1, op->tag, op->bytes, MPI_SUCCESS);
             }
-            if (it->bounce) (void)hipHostFree(it->bounce);
-            op->ch_done.store(1, std::memory_order_release);
+            free_bounce(*it);
             it = outstanding_.erase(it);
         }
     }
 
-    const char *name() const override { return "mpi-passthrough"; }
+    /* ------------------------------------------- partitioned: send side */
 
-private:
-    struct Out {
-        Op *op = nullptr;
-        MPI_Request req = MPI_REQUEST_NULL;
-        void *bounce = nullptr;
-        bool is_recv = false;
-    };
-    std::list<Out> outstanding_;
+    int start_psend_part(Op *op) {
+#if MPIX_HAVE_MPI_PARTITIONED
+        if (op->req != nullptr && op->req->mpi_part_native) {
+            /* reference init.cpp:82-86: Pready then mark completed; the
+             * persistent request itself completes in the host MPIX_Wait */
+            int rc = MPI_Pready(op->partition, op->req->mpi_preq);
+            if (rc != MPI_SUCCESS) return -1;
+            complete(op, -1, op->tag, op->bytes, MPI_SUCCESS);
+            return 0;
+        }
+#endif
+        /* emulation: header-prefixed message on the reserved tag */
+        Out o;
+        o.op = op;
+        o.is_part_send = true;
+        size_t n = sizeof(PartHdr) + op->bytes;
+        if (op->buf_is_device) {
+            /* pinned staging for the D2H copy */
+            if (hipHostMalloc(&o.bounce, n, 0) != hipSuccess) o.bounce = nullptr;
+            else o.bounce_pinned = true;
+        }
+        if (o.bounce == nullptr) o.bounce = malloc(n);
+        if (o.bounce == nullptr) {
+            MPIX_ERR("partition bounce alloc failed (%zu B)", n);
+            return -1;
+        }
+        PartHdr h;
+        h.partition = op->partition;
+        h.tag = op->tag;
+        h.pseq = op->pseq;
+        h.bytes = op->bytes;
+        memcpy(o.bounce, &h, sizeof(h));
+        if (op->bytes > 0) {
+            hipError_t e = op->buf_is_device
+                ? hipMemcpy((char *)o.bounce + sizeof(h), op->buf, op->bytes,
+                            hipMemcpyDeviceToHost)
+                : (memcpy((char *)o.bounce + sizeof(h), op->buf, op->bytes),
+                   hipSuccess);
+            if (e != hipSuccess) {
+                free_bounce(o);
+                return -1;
+            }
+        }
+        if (MPI_Isend(o.bounce, (int)n, MPI_BYTE, op->peer,
+                      reserved_part_tag(), op->comm, &o.req) != MPI_SUCCESS) {
+            free_bounce(o);
+            return -1;
+        }
+        outstanding_.push_back(o);
+        return 0;
+    }
+
+    /* ------------------------------------------- partitioned: recv side */
+
+    int start_precv_part(Op *op) {
+#if MPIX_HAVE_MPI_PARTITIONED
+        if (op->req != nullptr && op->req->mpi_part_native) {
+            parrived_.push_back(op);
+            return 0;
+        }
+#endif
+        /* early arrival already buffered? */
+        for (auto it = part_unexpected_.begin(); it != part_unexpected_.end();
+             ++it) {
+            if (it->comm == op->comm && it->src == op->peer &&
+                it->hdr.tag == op->tag && it->hdr.partition == op->partition &&
+                it->hdr.pseq == op->pseq) {
+                deliver_part(op, it->hdr, it->payload.data(), it->src);
+                part_unexpected_.erase(it);
+                return 0;
+            }
+        }
+        part_recvs_.push_back(op);
+        return 0;
+    }
+
+    void progress_parrived() {
+#if MPIX_HAVE_MPI_PARTITIONED
+        for (size_t i = 0; i < parrived_.size();) {
+            Op *op = parrived_[i];
+            int f = 0;
+            int rc = MPI_Parrived(op->req->mpi_preq, op->partition, &f);
+            if (rc != MPI_SUCCESS) {
+                complete(op, op->peer, op->tag, 0, MPI_ERR_OTHER);
+                f = 1;
+            } else if (f) {
+                complete(op, op->peer, op->tag, op->bytes, MPI_SUCCESS);
+            }
+            if (f) {
+                parrived_[i] = parrived_.back();
+                parrived_.pop_back();
+            } else {
+                i++;
+            }
+        }
+#endif
+    }
+
+    void deliver_part(Op *op, const PartHdr &h, const char *payload, int src) {
+        uint64_t n = h.bytes <= op->bytes ? h.bytes : op->bytes;
+        int err = h.bytes > op->bytes ? MPI_ERR_TRUNCATE : MPI_SUCCESS;
+        if (n > 0) {
+            hipError_t e = op->buf_is_device
+                ? hipMemcpy(op->buf, payload, n, hipMemcpyHostToDevice)
+                : (memcpy(op->buf, payload, n), hipSuccess);
+            if (e != hipSuccess) err = MPI_ERR_OTHER;
+        }
+        complete(op, src, h.tag, n, err);
+    }
+
+    void progress_part_probe() {
+        /* Probe only communicators with a pending partition recv: probing
+         * unconditionally would touch comms the user may already have freed
+         * (messages for not-yet-started requests just wait, correctly, in
+         * the MPI library's unexpected queue). */
+        if (part_recvs_.empty()) return;
+        std::vector<MPI_Comm> comms;
+        for (Op *op : part_recvs_) {
+            bool seen = false;
+            for (MPI_Comm c : comms)
+                if (c == op->comm) { seen = true; break; }
+            if (!seen) comms.push_back(op->comm);
+        }
+        for (MPI_Comm comm : comms) {
+            int flag = 0;
+            MPI_Status st;
+            while (MPI_Iprobe(MPI_ANY_SOURCE, reserved_part_tag(), comm,
+                              &flag, &st) == MPI_SUCCESS && flag) {
+                int n = 0;
+                MPI_Get_count(&st, MPI_BYTE, &n);
+                std::vector<char> buf((size_t)n);
+                if (MPI_Recv(buf.data(), n, MPI_BYTE, st.MPI_SOURCE,
+                             reserved_part_tag(), comm,
+                             MPI_STATUS_IGNORE) != MPI_SUCCESS)
+                    break;
+                if ((size_t)n < sizeof(PartHdr)) {
+                    MPIX_ERR("short partition message (%d B) from %d", n,
+                             st.MPI_SOURCE);
+                    continue;
+                }
+                PartHdr h;
+                memcpy(&h, buf.data(), sizeof(h));
+                if (h.magic != 0x4d505850u) {
+                    MPIX_ERR("bad partition magic from %d (user traffic on "
+                             "the reserved tag MPI_TAG_UB?)", st.MPI_SOURCE);
+                    continue;
+                }
+                bool matched = false;
+                for (size_t i = 0; i < part_recvs_.size(); i++) {
+                    Op *op = part_recvs_[i];
+                    if (op->comm == comm && op->peer == st.MPI_SOURCE &&
+                        op->tag == h.tag && op->partition == h.partition &&
+                        op->pseq == h.pseq) {
+                        deliver_part(op, h, buf.data() + sizeof(h),
+                                     st.MPI_SOURCE);
+                        part_recvs_.erase(part_recvs_.begin() + (long)i);
+                        matched = true;
+                        break;
+                    }
+                }
+                if (!matched) {
+                    PartMsg m;
+                    m.comm = comm;
+                    m.src = st.MPI_SOURCE;
+                    m.hdr = h;
+                    m.payload.assign(buf.begin() + sizeof(PartHdr), buf.end());
+                    part_unexpected_.push_back(std::move(m));
+                }
+            }
+        }
+    }
 };
 
 Transport *make_mpi_transport() { return new MpiTransport(); }

@@ -14,7 +14,7 @@ MPIEXEC = "/opt/conda/bin/mpiexec"
 
 ALL_TESTS = ["ring", "ring_all", "ring_all_device", "ring_all_graph",
              "ring_all_graph_construction", "ring_partitioned",
-             "ring_subcomm"]
+             "ring_subcomm", "ring_partitioned_subcomm"]
 
 
 def _ensure_built():
