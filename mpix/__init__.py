@@ -118,6 +118,11 @@ def waitall_graph(reqs):
     return _C.waitall_graph(list(reqs))
 
 
+def wait_graph(req):
+    """Per-request wait graph (explicit-construction composition)."""
+    return _C.wait_graph(req)
+
+
 def wait_enqueue(req, stream=None, status=None):
     _C.wait_enqueue(req, status, _stream_addr(stream))
 
@@ -181,6 +186,7 @@ def prequest_free(preq):
 # ----------------------------- graph helpers ------------------------------
 
 graph_chain_instantiate = _C.graph_chain_instantiate
+graph_instantiate = _C.graph_instantiate
 graph_launch = _C.graph_launch
 graph_exec_destroy = _C.graph_exec_destroy
 graph_destroy = _C.graph_destroy
