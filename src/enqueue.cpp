@@ -269,6 +269,7 @@ static int enqueue_sendrecv(bool is_send, void *buf, int count,
         }
     }
 
+    if (s->stats) op->t_enq_ns = now_ns();
     slot_arm(idx);
 
     int rc = fire_trigger(idx, qtype, queue, req);

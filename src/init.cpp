@@ -340,6 +340,15 @@ extern "C" int MPIX_Finalize(void)
                 fprintf(stderr, " [%d]=%lu", b,
                         (unsigned long)s->lat_hist[b]);
         fprintf(stderr, "\n");
+        if (s->leg_n) {
+            double k = 1e3 * (double)s->leg_n;
+            fprintf(stderr,
+                    "[mpix stats r%d] per-leg mean us over %lu ops: "
+                    "trigger->detect %.1f, transport %.1f, publish %.1f\n",
+                    s->world_rank, (unsigned long)s->leg_n,
+                    (double)s->leg_trig_ns / k, (double)s->leg_xfer_ns / k,
+                    (double)s->leg_compl_ns / k);
+        }
     }
 
     if (s->t_native) {

@@ -17,6 +17,7 @@
 #define MPIX_INTERNAL_H
 
 #include <atomic>
+#include <chrono>
 #include <cstdint>
 #include <memory>
 #include <cstdio>
@@ -96,6 +97,12 @@ namespace mpix {
 
 /* ------------------------------------------------------------- descriptors */
 
+static inline uint64_t now_ns()
+{
+    return (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
+        std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+
 struct Request;
 
 /* Transport-level completion record (filled by the channel). */
@@ -127,6 +134,7 @@ struct Op {
     Request *req = nullptr;         /* owning request */
 
     /* transport hand-off (written by channel, read by proxy) */
+    uint64_t t_enq_ns = 0;          /* host enqueue time (MPIX_STATS=1) */
     uint64_t t_issue_ns = 0;        /* proxy-side stats (MPIX_STATS=1) */
     std::atomic<int> ch_done{0};
     ChStatus ch_status;
@@ -149,6 +157,7 @@ struct Op {
         comm = MPI_COMM_NULL; comm_id = 0; buf_is_device = false;
         native_route = true;
         partition = -1; pseq = 0; req = nullptr;
+        t_enq_ns = 0; t_issue_ns = 0;
         ch_done.store(0, std::memory_order_relaxed);
         ch_status = ChStatus{}; ch_priv = nullptr;
         fast = false; waiter_owns_req = false;
@@ -304,6 +313,11 @@ struct State {
     bool stats = false;
     uint64_t lat_hist[20] = {0};
     uint64_t lat_sum_ns = 0;
+    /* per-leg sums (proxy thread only): enqueue->PENDING-detected (includes
+     * the GPU-side trigger when the stream is idle), detect->transport-done,
+     * done->completion-published */
+    uint64_t leg_trig_ns = 0, leg_xfer_ns = 0, leg_compl_ns = 0;
+    uint64_t leg_n = 0;
 };
 
 extern State *g_state;

@@ -19,14 +19,6 @@
  */
 #include <hip/hip_runtime.h>
 
-#include <chrono>
-
-static inline uint64_t now_ns()
-{
-    return (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
-        std::chrono::steady_clock::now().time_since_epoch()).count();
-}
-
 #include "internal.h"
 
 namespace mpix {
@@ -148,6 +140,11 @@ void proxy_main()
                  * completion state from the previous iteration */
                 op->ch_done.store(0, std::memory_order_relaxed);
                 op->status_saved = false;
+                if (s->stats && op->t_enq_ns) {
+                    uint64_t t = now_ns();
+                    s->leg_trig_ns += t - op->t_enq_ns;
+                    op->t_enq_ns = 0;
+                }
                 int rc = route(op)->start(op);
                 if (rc == 0) {
                     MPIX_TRACE_EV("slot %d %s peer=%d tag=%d part=%d bytes=%lu"
@@ -174,6 +171,16 @@ void proxy_main()
                     /* fast-wait completion: status into the request, seq
                      * word release-store (waiters use GTE), slot freed NOW
                      * — no COMPLETED/CLEANUP round trip */
+                    if (s->stats && op->t_issue_ns) {
+                        uint64_t d = now_ns() - op->t_issue_ns;
+                        s->lat_sum_ns += d;
+                        s->leg_xfer_ns += d;
+                        uint64_t us = d / 1000;
+                        int b = 0;
+                        while (us > 1 && b < 19) { us >>= 1; b++; }
+                        s->lat_hist[b]++;
+                        s->leg_n++;
+                    }
                     Request *req = op->req;
                     std::lock_guard<std::mutex> lk(s->completion_mutex);
                     fill_status(&req->fast_status, op->ch_status);
@@ -192,20 +199,26 @@ void proxy_main()
                 }
                 if (op->ch_done.load(std::memory_order_acquire)) {
                     if (s->stats && op->t_issue_ns) {
-                        uint64_t d = now_ns() - op->t_issue_ns;
+                        uint64_t t = now_ns();
+                        uint64_t d = t - op->t_issue_ns;
                         s->lat_sum_ns += d;
+                        s->leg_xfer_ns += d;
                         uint64_t us = d / 1000;
                         int b = 0;
                         while (us > 1 && b < 19) { us >>= 1; b++; }
                         s->lat_hist[b]++;
+                        s->leg_n++;
                     }
                     MPIX_TRACE_EV("slot %d ISSUED->COMPLETED err=%d", idx,
                                   op->ch_status.err);
+                    uint64_t tc = s->stats ? now_ns() : 0;
                     if (complete_op(idx, op)) { /* orphan: slot freed */
+                        if (s->stats) s->leg_compl_ns += now_ns() - tc;
                         drop(i);
                         did = true;
                         continue;
                     }
+                    if (s->stats) s->leg_compl_ns += now_ns() - tc;
                     did = true;
                 }
                 break;
