@@ -62,7 +62,9 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define BN 128
 #define BK 32
 #ifndef MPIX_GEMM_VARIANT
-#define MPIX_GEMM_VARIANT 2
+/* default = variant 3: BK=64 double-buffered LDS + register prefetch,
+ * 510.9 TF @8192^3 vs 476.1 (v2) / 236 (v1) — profiles/r02_gemm_v3_8k.json */
+#define MPIX_GEMM_VARIANT 3
 #endif
 
 #if MPIX_GEMM_VARIANT == 3

@@ -53,9 +53,9 @@ if [ "$MODE" = full ]; then
         && { echo OK; head -5 gpurun_out/devpush_pingpong.log; } \
         || echo "FAIL rc=$?"
     step gemm_8k 90 bench/bin/gemm_pready 8192 8192 8192 5
-    step gemm_v3_check 90 bench/bin/gemm_pready_v3 --check
-    step gemm_v3_8k 90 bench/bin/gemm_pready_v3 8192 8192 8192 5
-    tail -1 gpurun_out/gemm_v3_8k.log
+    step gemm_v2_check 90 bench/bin/gemm_pready_v2 --check
+    step gemm_v2_8k 90 bench/bin/gemm_pready_v2 8192 8192 8192 5
+    tail -1 gpurun_out/gemm_v2_8k.log
     note rocprof
     ( cd /tmp && timeout 150 rocprofv3 --kernel-trace --stats \
         -d "$OLDPWD/gpurun_out/prof" -o ci \
