@@ -272,6 +272,36 @@ extern "C" int MPIX_Init(void)
     s->proxy_stop.store(false);
     s->proxy = std::thread(proxy_main);
 
+    /* MPIX_WATCHDOG=<secs>: periodic flag-pool dump for hang diagnosis.
+     * Detached-by-join at finalize via proxy_stop. */
+    if (int wd = env_int("MPIX_WATCHDOG", 0)) {
+        s->watchdog = std::thread([s, wd] {
+            int tick = 0;
+            while (!s->proxy_stop.load(std::memory_order_acquire)) {
+                for (int ms = 0; ms < wd * 1000 &&
+                     !s->proxy_stop.load(std::memory_order_acquire); ms += 100)
+                    std::this_thread::sleep_for(std::chrono::milliseconds(100));
+                if (s->proxy_stop.load(std::memory_order_acquire)) break;
+                fprintf(stderr, "[mpix watchdog r%d t+%ds] active slots:",
+                        s->world_rank, ++tick * wd);
+                for (size_t i = 0; i < s->nflags; i++) {
+                    uint32_t f = s->flags[i].load(std::memory_order_relaxed);
+                    if (f != MPIX_FLAG_AVAILABLE)
+                        fprintf(stderr, " [%zu]=%u(kind%d,peer%d,tag%d,done%d)",
+                                i, f, (int)s->ops[i].kind, s->ops[i].peer_world,
+                                s->ops[i].tag,
+                                s->ops[i].ch_done.load(
+                                    std::memory_order_relaxed));
+                }
+                fprintf(stderr, " issued=%lu completed=%lu\n",
+                        (unsigned long)s->ops_issued.load(
+                            std::memory_order_relaxed),
+                        (unsigned long)s->ops_completed.load(
+                            std::memory_order_relaxed));
+            }
+        });
+    }
+
     MPIX_DBG("init done: rank %d/%d mpi_mode=%d gpu=%d dev=%d memops=%d batch=%d "
              "nflags=%zu", s->world_rank, s->world_size, s->mpi_mode,
              s->have_gpu, s->device_id, s->use_memops, s->use_batch_memops,
@@ -300,6 +330,7 @@ extern "C" int MPIX_Finalize(void)
 
     s->proxy_stop.store(true);
     if (s->proxy.joinable()) s->proxy.join();
+    if (s->watchdog.joinable()) s->watchdog.join();
 
     /* From here on, state teardown must exclude late hipUserObject
      * destructors (graph_request_destroy): they check g_state + generation

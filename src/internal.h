@@ -297,6 +297,7 @@ struct State {
     std::atomic<uint32_t> alloc_cursor{0};
     /* proxy */
     std::thread proxy;
+    std::thread watchdog;            /* MPIX_WATCHDOG debug dumper */
     std::atomic<bool> proxy_stop{false};
     SlotRing armed;                          /* slots newly allocated */
     std::mutex completion_mutex;             /* status-delivery race guard */
