@@ -37,16 +37,26 @@ step gemm_check 90 bench/bin/gemm_pready --check
 step gemm_overlap 120 mpiexec -np 2 bench/bin/gemm_pready 4096 4096 4096 5
 
 if [ "$MODE" = full ]; then
-    # fast-wait A/B (epoch/GTE wait protocol) — round-2 measurement
-    note fastwait_pytest
-    MPIX_FAST_WAIT=1 timeout 180 python -m pytest tests/test_gpu.py -q -m gpu \
-        -k "loopback or partitioned" > gpurun_out/fastwait_pytest.log 2>&1 \
-        && echo OK || { echo "FAIL rc=$?"; tail -4 gpurun_out/fastwait_pytest.log; }
-    note fastwait_pingpong
-    MPIX_FAST_WAIT=1 timeout 100 mpiexec -np 2 bench/bin/pingpong 18 100 \
-        > gpurun_out/fastwait_pingpong.log 2>&1 \
-        && { echo OK; head -6 gpurun_out/fastwait_pingpong.log; } \
+    # classic-protocol A/B (fast-wait is the default since r02)
+    note classic_pytest
+    MPIX_FAST_WAIT=0 timeout 240 python -m pytest tests/test_gpu.py -q -m gpu \
+        -k "loopback or partitioned" > gpurun_out/classic_pytest.log 2>&1 \
+        && echo OK || { echo "FAIL rc=$?"; tail -4 gpurun_out/classic_pytest.log; }
+    note classic_pingpong
+    MPIX_FAST_WAIT=0 timeout 100 mpiexec -np 2 bench/bin/pingpong 18 100 \
+        > gpurun_out/classic_pingpong.log 2>&1 \
+        && { echo OK; head -6 gpurun_out/classic_pingpong.log; } \
         || echo "FAIL rc=$?"
+    note stats_pingpong
+    MPIX_STATS=1 timeout 100 mpiexec -np 2 bench/bin/pingpong 14 200 \
+        > gpurun_out/stats_pingpong.log 2>&1 \
+        && { echo OK; grep "per-leg" gpurun_out/stats_pingpong.log; } \
+        || echo "FAIL rc=$?"
+    note flush_probe
+    ( timeout 60 tools/bin/flush_probe 1 && timeout 60 tools/bin/flush_probe 0 ) \
+        > gpurun_out/flush_probe.log 2>&1 \
+        && { echo OK; cat gpurun_out/flush_probe.log; } \
+        || { echo "FAIL rc=$?"; cat gpurun_out/flush_probe.log; }
     note devpush_pingpong
     MPIX_DEV_PUSH_MAX=65536 timeout 100 mpiexec -np 2 bench/bin/pingpong 15 50 \
         > gpurun_out/devpush_pingpong.log 2>&1 \
