@@ -36,9 +36,10 @@ def test_pool_exhaustion_and_recovery(mpix_small_pool):
             reqs.append(mpix.isend_enqueue(bufs[i % n], dest=0, tag=i))
     except RuntimeError:
         raised = True
-    if os.environ.get("MPIX_FAST_WAIT") == "1":
-        # fast-wait frees slots the moment the proxy completes a send, so
-        # the pool may never exhaust here — only the drain below is checked
+    if os.environ.get("MPIX_FAST_WAIT", "1") != "0":
+        # fast-wait (the default) frees slots the moment the proxy completes
+        # a send, so the pool may never exhaust here — only the drain below
+        # is checked
         pass
     else:
         assert raised, "expected pool exhaustion error"
