@@ -133,28 +133,77 @@ static int attach_cleanup(hipGraph_t graph, Request *req)
     return 0;
 }
 
-/* Build a single-kernel-node graph calling k_set_flag / k_wait_flag. */
-static int make_flag_graph(hipGraph_t *out, void *kernel, uint32_t *flag_d,
-                           uint32_t val)
+/* Add a flag trigger/wait node to `g`.  When the graph batch-memOp node is
+ * functional (probed at init) the node is a memory operation — crucial:
+ * memOp waits park in the queue scheduler instead of occupying a compute
+ * queue with a spinning kernel.  A spin-wait kernel node can deadlock the
+ * whole protocol when HIP's stream→HSA-queue multiplexing lands it on the
+ * same hardware queue as the transport's copy work (graph execution
+ * serializes queues with AQL barrier packets) — observed deterministically
+ * after a few stream-create cycles (gpurun_out/diag2_*).  The kernel-node
+ * fallback remains for memOps-less configs and carries that residual
+ * hazard (mitigate with GPU_MAX_HW_QUEUES; the transport's own copy
+ * stream is already on a dedicated CU-masked queue). */
+static int add_flag_node(hipGraph_t g, bool is_wait, uint32_t *flag_d,
+                         uint32_t val)
 {
-    hipGraph_t g = nullptr;
-    MPIX_CHECK_HIP(hipGraphCreate(&g, 0));
+    State *s = g_state;
+    hipGraphNode_t node = nullptr;
+    if (s->use_graph_memops) {
+        hipStreamBatchMemOpParams p;
+        memset(&p, 0, sizeof(p));
+        if (is_wait) {
+            p.waitValue.operation = hipStreamMemOpWaitValue32;
+            p.waitValue.address = flag_d;
+            p.waitValue.value = val;
+            p.waitValue.flags = hipStreamWaitValueEq;
+        } else {
+            p.writeValue.operation = hipStreamMemOpWriteValue32;
+            p.writeValue.address = flag_d;
+            p.writeValue.value = val;
+        }
+        hipBatchMemOpNodeParams np;
+        memset(&np, 0, sizeof(np));
+        np.ctx = nullptr; /* current context */
+        np.count = 1;
+        np.paramArray = &p;
+        np.flags = 0;
+        hipError_t e = hipGraphAddBatchMemOpNode(&node, g, nullptr, 0, &np);
+        if (e != hipSuccess) {
+            MPIX_ERR("hipGraphAddBatchMemOpNode failed: %s",
+                     hipGetErrorString(e));
+            return (int)e;
+        }
+        return 0;
+    }
     hipKernelNodeParams p{};
     /* hipGraphAddKernelNode copies parameter values during the call, so
      * locals are fine here */
     void *kp[2] = {&flag_d, &val};
-    p.func = kernel;
+    p.func = is_wait ? (void *)k_wait_flag : (void *)k_set_flag;
     p.gridDim = dim3(1, 1, 1);
     p.blockDim = dim3(1, 1, 1);
     p.sharedMemBytes = 0;
     p.kernelParams = kp;
     p.extra = nullptr;
-    hipGraphNode_t node = nullptr;
     hipError_t e = hipGraphAddKernelNode(&node, g, nullptr, 0, &p);
     if (e != hipSuccess) {
         MPIX_ERR("hipGraphAddKernelNode failed: %s", hipGetErrorString(e));
-        (void)hipGraphDestroy(g);
         return (int)e;
+    }
+    return 0;
+}
+
+/* Build a single-node graph triggering or waiting a flag. */
+static int make_flag_graph(hipGraph_t *out, bool is_wait, uint32_t *flag_d,
+                           uint32_t val)
+{
+    hipGraph_t g = nullptr;
+    MPIX_CHECK_HIP(hipGraphCreate(&g, 0));
+    int rc = add_flag_node(g, is_wait, flag_d, val);
+    if (rc != 0) {
+        (void)hipGraphDestroy(g);
+        return rc;
     }
     *out = g;
     return 0;
@@ -173,8 +222,7 @@ static int fire_trigger(int idx, int qtype, void *queue, Request *req)
             return MPI_ERR_OTHER;
         }
         hipGraph_t g = nullptr;
-        MPIX_CHECK(make_flag_graph(&g, (void *)k_set_flag, flag_d,
-                                   MPIX_FLAG_PENDING));
+        MPIX_CHECK(make_flag_graph(&g, false, flag_d, MPIX_FLAG_PENDING));
         MPIX_CHECK(attach_cleanup(g, req));
         *(hipGraph_t *)queue = g;
         return MPI_SUCCESS;
@@ -187,9 +235,14 @@ static int fire_trigger(int idx, int qtype, void *queue, Request *req)
     }
     hipStream_t stream = queue ? *(hipStream_t *)queue : (hipStream_t)0;
     if (stream_capturing(stream)) {
-        hipLaunchKernelGGL(k_set_flag, dim3(1), dim3(1), 0, stream, flag_d,
-                           (uint32_t)MPIX_FLAG_PENDING);
-        MPIX_CHECK_HIP(hipGetLastError());
+        if (s->use_capture_memops) {
+            MPIX_CHECK_HIP(hipStreamWriteValue32(
+                stream, flag_d, (uint32_t)MPIX_FLAG_PENDING, 0));
+        } else {
+            hipLaunchKernelGGL(k_set_flag, dim3(1), dim3(1), 0, stream,
+                               flag_d, (uint32_t)MPIX_FLAG_PENDING);
+            MPIX_CHECK_HIP(hipGetLastError());
+        }
         /* request lifetime follows the captured graph */
         hipStreamCaptureStatus cst;
         unsigned long long cid = 0;
@@ -397,8 +450,7 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
         }
         hipGraph_t g = nullptr;
         /* graph wait targets COMPLETED (reference bug D2 fixed) */
-        MPIX_CHECK(make_flag_graph(&g, (void *)k_wait_flag, flag_d,
-                                   MPIX_FLAG_COMPLETED));
+        MPIX_CHECK(make_flag_graph(&g, true, flag_d, MPIX_FLAG_COMPLETED));
         *(hipGraph_t *)queue = g;
         *reqp = MPIX_REQUEST_NULL; /* ownership: send/recv graph's user object */
         return MPI_SUCCESS;
@@ -417,9 +469,18 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
             std::lock_guard<std::mutex> lk(s->completion_mutex);
             s->ops[idx].enq_status_target = status;
         }
-        hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream, flag_d,
-                           (uint32_t)MPIX_FLAG_COMPLETED);
-        MPIX_CHECK_HIP(hipGetLastError());
+        /* poll-only wait (no CLEANUP write): the captured graph relaunches
+         * and recycles the flag; cleanup rides the graph user object */
+        if (s->use_capture_memops) {
+            MPIX_CHECK_HIP(hipStreamWaitValue32(stream, flag_d,
+                                                (uint32_t)MPIX_FLAG_COMPLETED,
+                                                hipStreamWaitValueEq,
+                                                0xFFFFFFFFu));
+        } else {
+            hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
+                               flag_d, (uint32_t)MPIX_FLAG_COMPLETED);
+            MPIX_CHECK_HIP(hipGetLastError());
+        }
         *reqp = MPIX_REQUEST_NULL;
         return MPI_SUCCESS;
     }
@@ -482,19 +543,11 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
                         s->ops[idx].enq_status_target = st;
                 }
             }
-            hipKernelNodeParams p{};
-            uint32_t *flag_d = s->flags_d + idx;
-            uint32_t val = MPIX_FLAG_COMPLETED;
-            void *kp[2] = {&flag_d, &val};
-            p.func = (void *)k_wait_flag;
-            p.gridDim = dim3(1, 1, 1);
-            p.blockDim = dim3(1, 1, 1);
-            p.kernelParams = kp;
-            hipGraphNode_t node = nullptr;
-            hipError_t e = hipGraphAddKernelNode(&node, g, nullptr, 0, &p);
-            if (e != hipSuccess) {
+            int arc = add_flag_node(g, true, s->flags_d + idx,
+                                    MPIX_FLAG_COMPLETED);
+            if (arc != 0) {
                 (void)hipGraphDestroy(g);
-                return (int)e;
+                return arc;
             }
             reqs[i] = MPIX_REQUEST_NULL;
         }
@@ -601,10 +654,16 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
                 MPI_Status *st = status_at(i);
                 if (st) s->ops[idx].enq_status_target = st;
             }
-            hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
-                               s->flags_d + idx,
-                               (uint32_t)MPIX_FLAG_COMPLETED);
-            MPIX_CHECK_HIP(hipGetLastError());
+            if (s->use_capture_memops) {
+                MPIX_CHECK_HIP(hipStreamWaitValue32(
+                    stream, s->flags_d + idx, (uint32_t)MPIX_FLAG_COMPLETED,
+                    hipStreamWaitValueEq, 0xFFFFFFFFu));
+            } else {
+                hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
+                                   s->flags_d + idx,
+                                   (uint32_t)MPIX_FLAG_COMPLETED);
+                MPIX_CHECK_HIP(hipGetLastError());
+            }
             reqs[i] = MPIX_REQUEST_NULL;
         }
         return MPI_SUCCESS;

@@ -129,6 +129,166 @@ static void destroy_pools(State *s)
 
 static std::atomic<uint64_t> g_init_gen{0};
 
+/* Probe hipGraphAddBatchMemOpNode semantics: write node fires, wait node
+ * (a) passes once its value is published and (b) does NOT pass before.
+ * Needed because graph-node spin-wait KERNELS can deadlock against other
+ * work via HIP's stream→HSA-queue multiplexing (see add_flag_node). */
+static void probe_graph_memops(State *s)
+{
+    s->use_graph_memops = false;
+    if (!s->have_gpu || !s->use_memops) return;
+    if (env_int("MPIX_DISABLE_GRAPH_MEMOPS", 0)) return;
+    const int idx = (int)s->nflags - 1;
+    uint32_t *fd = s->flags_d + idx;
+    const uint32_t magic = 0x6D700000u;
+    hipStream_t st = nullptr;
+    if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) != hipSuccess) {
+        (void)hipGetLastError();
+        return;
+    }
+    bool ok = false;
+    hipGraph_t g = nullptr, g2 = nullptr;
+    hipGraphExec_t ge = nullptr, ge2 = nullptr;
+    do {
+        s->flags[idx].store(0);
+        if (hipGraphCreate(&g, 0) != hipSuccess) break;
+        auto op_write = [&](uint32_t v) {
+            hipStreamBatchMemOpParams p;
+            memset(&p, 0, sizeof(p));
+            p.writeValue.operation = hipStreamMemOpWriteValue32;
+            p.writeValue.address = fd;
+            p.writeValue.value = v;
+            return p;
+        };
+        auto op_wait = [&](uint32_t v) {
+            hipStreamBatchMemOpParams p;
+            memset(&p, 0, sizeof(p));
+            p.waitValue.operation = hipStreamMemOpWaitValue32;
+            p.waitValue.address = fd;
+            p.waitValue.value = v;
+            p.waitValue.flags = hipStreamWaitValueEq;
+            return p;
+        };
+        hipStreamBatchMemOpParams pw = op_write(magic), pq = op_wait(magic),
+                                  pw2 = op_write(magic + 1);
+        hipBatchMemOpNodeParams np;
+        memset(&np, 0, sizeof(np));
+        np.count = 1;
+        hipGraphNode_t n1, n2, n3;
+        np.paramArray = &pw;
+        if (hipGraphAddBatchMemOpNode(&n1, g, nullptr, 0, &np) != hipSuccess)
+            break;
+        np.paramArray = &pq;
+        if (hipGraphAddBatchMemOpNode(&n2, g, &n1, 1, &np) != hipSuccess)
+            break;
+        np.paramArray = &pw2;
+        if (hipGraphAddBatchMemOpNode(&n3, g, &n2, 1, &np) != hipSuccess)
+            break;
+        if (hipGraphInstantiate(&ge, g, nullptr, nullptr, 0) != hipSuccess)
+            break;
+        if (hipGraphLaunch(ge, st) != hipSuccess) break;
+        bool done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        (void)hipGetLastError();
+        if (!done || s->flags[idx].load() != magic + 1) break;
+
+        /* unsatisfied wait must BLOCK until the host publishes the value */
+        if (hipGraphCreate(&g2, 0) != hipSuccess) break;
+        hipStreamBatchMemOpParams pq2 = op_wait(magic + 2);
+        np.paramArray = &pq2;
+        hipGraphNode_t m1;
+        if (hipGraphAddBatchMemOpNode(&m1, g2, nullptr, 0, &np) != hipSuccess)
+            break;
+        if (hipGraphInstantiate(&ge2, g2, nullptr, nullptr, 0) != hipSuccess)
+            break;
+        if (hipGraphLaunch(ge2, st) != hipSuccess) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+        if (hipStreamQuery(st) != hipErrorNotReady) {
+            /* no-op wait: passed without its value — unusable */
+            (void)hipGetLastError();
+            break;
+        }
+        s->flags[idx].store(magic + 2, std::memory_order_release);
+        done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        (void)hipGetLastError();
+        if (!done) {
+            fprintf(stderr, "[mpix] warn: graph memOp wait never observed a "
+                    "host store; leaving graph memOps off\n");
+            break;
+        }
+        ok = true;
+    } while (0);
+    (void)hipGetLastError();
+    if (ge) (void)hipGraphExecDestroy(ge);
+    if (ge2) (void)hipGraphExecDestroy(ge2);
+    if (g) (void)hipGraphDestroy(g);
+    if (g2) (void)hipGraphDestroy(g2);
+    (void)hipStreamDestroy(st); /* deferred if a packet is stuck */
+    s->flags[idx].store(MPIX_FLAG_AVAILABLE);
+    s->use_graph_memops = ok;
+}
+
+/* Probe whether WriteValue32/WaitValue32 can be RECORDED during stream
+ * capture (replaces the captured spin-wait kernels when possible). */
+static void probe_capture_memops(State *s)
+{
+    s->use_capture_memops = false;
+    if (!s->have_gpu || !s->use_memops) return;
+    if (env_int("MPIX_DISABLE_GRAPH_MEMOPS", 0)) return;
+    const int idx = (int)s->nflags - 1;
+    uint32_t *fd = s->flags_d + idx;
+    const uint32_t magic = 0x63617000u;
+    hipStream_t st = nullptr;
+    if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) != hipSuccess) {
+        (void)hipGetLastError();
+        return;
+    }
+    bool ok = false;
+    hipGraph_t g = nullptr;
+    hipGraphExec_t ge = nullptr;
+    do {
+        s->flags[idx].store(0);
+        if (hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) !=
+            hipSuccess)
+            break;
+        bool rec_ok =
+            hipStreamWriteValue32(st, fd, magic, 0) == hipSuccess &&
+            hipStreamWaitValue32(st, fd, magic, hipStreamWaitValueEq,
+                                 0xFFFFFFFFu) == hipSuccess;
+        if (hipStreamEndCapture(st, &g) != hipSuccess || !rec_ok || !g)
+            break;
+        if (hipGraphInstantiate(&ge, g, nullptr, nullptr, 0) != hipSuccess)
+            break;
+        if (hipGraphLaunch(ge, st) != hipSuccess) break;
+        bool done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        if (!done || s->flags[idx].load() != magic) break;
+        ok = true;
+    } while (0);
+    (void)hipGetLastError();
+    if (ge) (void)hipGraphExecDestroy(ge);
+    if (g) (void)hipGraphDestroy(g);
+    (void)hipStreamDestroy(st);
+    s->flags[idx].store(MPIX_FLAG_AVAILABLE);
+    s->use_capture_memops = ok;
+}
+
 extern "C" int MPIX_Init(void)
 {
     if (g_state != nullptr) {
@@ -259,6 +419,12 @@ extern "C" int MPIX_Init(void)
 
     /* --- memOps fast path ---------------------------------------------- */
     probe_memops(s);
+    probe_graph_memops(s);
+    probe_capture_memops(s);
+    if (s->have_gpu)
+        MPIX_DBG("probes: memops=%d batch=%d graph=%d capture=%d",
+                 s->use_memops, s->use_batch_memops, s->use_graph_memops,
+                 s->use_capture_memops);
 
     /* --- data plane ----------------------------------------------------- */
     s->t_native = make_native_transport(s->world_rank, s->world_size,

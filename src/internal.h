@@ -283,6 +283,8 @@ struct State {
     int device_id = -1;
     bool use_memops = false;    /* hipStreamWriteValue32/WaitValue32 path */
     bool use_batch_memops = false;
+    bool use_graph_memops = false; /* hipGraphAddBatchMemOpNode functional */
+    bool use_capture_memops = false; /* memOps recordable under capture */
     /* flag pool */
     size_t nflags = 0;
     std::atomic<uint32_t> *flags = nullptr;  /* host view (pinned if GPU) */

@@ -354,8 +354,32 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
-            hipSuccess) {
+        /* The copy stream MUST own a dedicated hardware queue.  HIP muxes
+         * ordinary streams onto a small pool of HSA queues (4 by default),
+         * and graph execution serializes a queue with AQL barrier packets:
+         * when a user graph's spin-wait kernel (k_wait_flag, waiting for a
+         * flag only THIS stream's pull-copy can eventually satisfy) landed
+         * on the same queue as the copy stream, the copy could never start
+         * — a deterministic deadlock after enough stream-create cycles
+         * rotated the mapping (gpurun_out/diag2_*: slots stuck ISSUED,
+         * ch_done=0).  CU-masked streams get their own queue; mask = all
+         * CUs so there is no compute restriction. */
+        hipDeviceProp_t prop{};
+        int ncu = 0;
+        if (hipGetDeviceProperties(&prop, dev_) == hipSuccess)
+            ncu = prop.multiProcessorCount;
+        if (ncu > 0) {
+            std::vector<uint32_t> mask((size_t)(ncu + 31) / 32, 0xFFFFFFFFu);
+            if (hipExtStreamCreateWithCUMask(&copy_stream_,
+                                             (uint32_t)mask.size(),
+                                             mask.data()) != hipSuccess) {
+                (void)hipGetLastError();
+                copy_stream_ = nullptr;
+            }
+        }
+        if (copy_stream_ == nullptr &&
+            hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
+                hipSuccess) {
             MPIX_ERR("copy stream create failed");
             return -1;
         }
