@@ -52,6 +52,19 @@ if [ "$MODE" = full ]; then
         > gpurun_out/stats_pingpong.log 2>&1 \
         && { echo OK; grep "per-leg" gpurun_out/stats_pingpong.log; } \
         || echo "FAIL rc=$?"
+    note waitall_wave_kernel
+    # single-wave waitall kernel under the kernel fallback: the r01 in-situ
+    # hang was the stream->HSA-queue aliasing deadlock (see add_flag_node);
+    # with the transport copy stream on its own queue this should pass
+    MPIX_WAITALL_KERNEL=1 MPIX_DISABLE_MEMOPS=1 timeout 120 \
+        python -m pytest tests/test_gpu.py -q -m gpu -k "waitall or loopback" \
+        > gpurun_out/waitall_wave.log 2>&1 \
+        && echo OK || { echo "FAIL rc=$?"; tail -4 gpurun_out/waitall_wave.log; }
+    note kernelwait_pingpong
+    MPIX_DISABLE_MEMOPS=1 timeout 100 mpiexec -np 2 bench/bin/pingpong 15 100 \
+        > gpurun_out/kernelwait_pingpong.log 2>&1 \
+        && { echo OK; head -5 gpurun_out/kernelwait_pingpong.log; } \
+        || echo "FAIL rc=$?"
     note flush_probe
     ( timeout 60 tools/bin/flush_probe 1 && timeout 60 tools/bin/flush_probe 0 ) \
         > gpurun_out/flush_probe.log 2>&1 \

@@ -253,3 +253,10 @@ def _partitioned_ring_device(rank, size):
 
 def test_partitioned_ring_device_2proc():
     run_ranks(2, _partitioned_ring_device, timeout=240)
+
+
+def test_ring_device_4proc_oversubscribed():
+    """4 ranks on however many GPUs the box has (1 on CI): exercises
+    multi-peer IPC handle exchange and per-pair shm rings — the shape the
+    driver's 8-GPU scale run hits with real xGMI peers."""
+    run_ranks(4, _ring_device_stream, timeout=240)
