@@ -421,10 +421,11 @@ extern "C" int MPIX_Init(void)
     probe_memops(s);
     probe_graph_memops(s);
     probe_capture_memops(s);
-    if (s->have_gpu)
-        MPIX_DBG("probes: memops=%d batch=%d graph=%d capture=%d",
-                 s->use_memops, s->use_batch_memops, s->use_graph_memops,
-                 s->use_capture_memops);
+    if (s->have_gpu && env_int("MPIX_TRACE", 0))
+        fprintf(stderr,
+                "[mpix r%d] probes: memops=%d batch=%d graph=%d capture=%d\n",
+                s->world_rank, s->use_memops, s->use_batch_memops,
+                s->use_graph_memops, s->use_capture_memops);
 
     /* --- data plane ----------------------------------------------------- */
     s->t_native = make_native_transport(s->world_rank, s->world_size,

@@ -285,6 +285,11 @@ struct State {
     bool use_batch_memops = false;
     bool use_graph_memops = false; /* hipGraphAddBatchMemOpNode functional */
     bool use_capture_memops = false; /* memOps recordable under capture */
+    /* sticky: a spin-WAIT kernel has been emitted (graph node, captured, or
+     * stream fallback).  While false, no spinning kernel can exist on any
+     * HSA queue, so the transport may run its pull copy as a compute
+     * kernel; once true, copies ride SDMA (see pull_kernels_safe). */
+    std::atomic<bool> spin_wait_kernels{false};
     /* flag pool */
     size_t nflags = 0;
     std::atomic<uint32_t> *flags = nullptr;  /* host view (pinned if GPU) */

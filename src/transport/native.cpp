@@ -354,37 +354,33 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        /* The copy stream MUST own a dedicated hardware queue.  HIP muxes
-         * ordinary streams onto a small pool of HSA queues (4 by default),
-         * and graph execution serializes a queue with AQL barrier packets:
-         * when a user graph's spin-wait kernel (k_wait_flag, waiting for a
-         * flag only THIS stream's pull-copy can eventually satisfy) landed
-         * on the same queue as the copy stream, the copy could never start
-         * — a deterministic deadlock after enough stream-create cycles
-         * rotated the mapping (gpurun_out/diag2_*: slots stuck ISSUED,
-         * ch_done=0).  CU-masked streams get their own queue; mask = all
-         * CUs so there is no compute restriction. */
-        hipDeviceProp_t prop{};
-        int ncu = 0;
-        if (hipGetDeviceProperties(&prop, dev_) == hipSuccess)
-            ncu = prop.multiProcessorCount;
-        if (ncu > 0) {
-            std::vector<uint32_t> mask((size_t)(ncu + 31) / 32, 0xFFFFFFFFu);
-            if (hipExtStreamCreateWithCUMask(&copy_stream_,
-                                             (uint32_t)mask.size(),
-                                             mask.data()) != hipSuccess) {
-                (void)hipGetLastError();
-                copy_stream_ = nullptr;
-            }
-        }
-        if (copy_stream_ == nullptr &&
-            hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
-                hipSuccess) {
+        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
+            hipSuccess) {
             MPIX_ERR("copy stream create failed");
             return -1;
         }
     }
     return 0;
+}
+
+/* May the transport run its pull copy as a COMPUTE KERNEL?  HIP muxes
+ * streams onto a small HSA hardware-queue pool and serializes a queue with
+ * AQL barrier packets during graph execution, so a spin-wait kernel (a
+ * k_wait_flag graph node, a captured wait, or the memOps-less stream-wait
+ * fallback) that lands on the same queue as the pull-copy kernel blocks it
+ * forever — and the spin waits for a flag only that copy can complete.
+ * Observed deterministically once enough stream-create cycles rotated the
+ * stream→queue mapping (gpurun_out/diag2_*: slots stuck ISSUED,ch_done=0;
+ * an attempted dedicated CU-masked queue for the copy stream wedged the
+ * runtime outright, diag3).  Deterministic rule instead: use the pull
+ * KERNEL only when every wait in the library is a memory operation
+ * (probed at init) — then no spin kernel can exist in front of it;
+ * otherwise ride SDMA, which compute-queue ordering cannot block. */
+static bool pull_kernels_safe()
+{
+    State *s = g_state;
+    return s != nullptr && s->use_memops && s->use_graph_memops &&
+           s->use_capture_memops;
 }
 
 void NativeTransport::shutdown()
@@ -777,7 +773,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * and (b) 16-byte alignment on both sides (user buffers can be
      * arbitrarily offset, e.g. partitioned slices).  Everything else rides
      * hipMemcpyAsync (SDMA / runtime blit). */
-    bool kernel_ok = op->buf_is_device &&
+    bool kernel_ok = op->buf_is_device && pull_kernels_safe() &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
     if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
