@@ -78,6 +78,15 @@ __global__ void k_waitall_and_set(uint32_t *flags, const int32_t *idx,
 
 /* ------------------------------------------------------------------ helpers */
 
+/* Record that a spin-wait kernel has been emitted somewhere (stream
+ * fallback, capture, or graph node) — flips the transport to SDMA copies
+ * so no copy kernel can ever be queued behind a spinning wave (see
+ * pull_kernels_safe in transport/native.cpp). */
+static inline void mark_spin_wait()
+{
+    g_state->spin_wait_kernels.store(true, std::memory_order_release);
+}
+
 static bool stream_capturing(hipStream_t stream)
 {
     hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
@@ -141,9 +150,9 @@ static int attach_cleanup(hipGraph_t graph, Request *req)
  * same hardware queue as the transport's copy work (graph execution
  * serializes queues with AQL barrier packets) — observed deterministically
  * after a few stream-create cycles (gpurun_out/diag2_*).  The kernel-node
- * fallback remains for memOps-less configs and carries that residual
- * hazard (mitigate with GPU_MAX_HW_QUEUES; the transport's own copy
- * stream is already on a dedicated CU-masked queue). */
+ * fallback remains for memOps-less configs; emitting one flips the
+ * transport to SDMA-only copies (mark_spin_wait) so the blocked-copy cycle
+ * cannot form. */
 static int add_flag_node(hipGraph_t g, bool is_wait, uint32_t *flag_d,
                          uint32_t val)
 {
@@ -176,6 +185,7 @@ static int add_flag_node(hipGraph_t g, bool is_wait, uint32_t *flag_d,
         }
         return 0;
     }
+    if (is_wait) mark_spin_wait();
     hipKernelNodeParams p{};
     /* hipGraphAddKernelNode copies parameter values during the call, so
      * locals are fine here */
@@ -425,6 +435,7 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
                                                 hipStreamWaitValueGte,
                                                 0xFFFFFFFFu));
         } else {
+            mark_spin_wait();
             hipLaunchKernelGGL(k_wait_flag_gte, dim3(1), dim3(1), 0, stream,
                                seq_d, wait_seq);
             MPIX_CHECK_HIP(hipGetLastError());
@@ -477,6 +488,7 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
                                                 hipStreamWaitValueEq,
                                                 0xFFFFFFFFu));
         } else {
+            mark_spin_wait();
             hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
                                flag_d, (uint32_t)MPIX_FLAG_COMPLETED);
             MPIX_CHECK_HIP(hipGetLastError());
@@ -496,7 +508,8 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
         MPIX_CHECK_HIP(hipStreamWriteValue32(stream, flag_d,
                                              (uint32_t)MPIX_FLAG_CLEANUP, 0));
     } else {
-        hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0, stream, flag_d,
+        mark_spin_wait();
+            hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0, stream, flag_d,
                            (uint32_t)MPIX_FLAG_COMPLETED,
                            (uint32_t)MPIX_FLAG_CLEANUP);
         MPIX_CHECK_HIP(hipGetLastError());
@@ -659,7 +672,8 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
                     stream, s->flags_d + idx, (uint32_t)MPIX_FLAG_COMPLETED,
                     hipStreamWaitValueEq, 0xFFFFFFFFu));
             } else {
-                hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
+                mark_spin_wait();
+            hipLaunchKernelGGL(k_wait_flag, dim3(1), dim3(1), 0, stream,
                                    s->flags_d + idx,
                                    (uint32_t)MPIX_FLAG_COMPLETED);
                 MPIX_CHECK_HIP(hipGetLastError());
@@ -691,7 +705,8 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
                 continue;
             }
             if (!try_complete_now(idx, status_at(i))) {
-                hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0,
+                mark_spin_wait();
+            hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0,
                                    stream, s->flags_d + idx,
                                    (uint32_t)MPIX_FLAG_COMPLETED,
                                    (uint32_t)MPIX_FLAG_CLEANUP);
@@ -740,7 +755,8 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         MPIX_CHECK_HIP(hipHostGetDevicePointer((void **)&idx_d, idx_arr, 0));
         int threads = n < 64 ? 64 : ((n + 63) / 64) * 64;
         if (threads > 1024) threads = 1024;
-        hipLaunchKernelGGL(k_waitall_and_set, dim3(1), dim3(threads), 0,
+        mark_spin_wait();
+            hipLaunchKernelGGL(k_waitall_and_set, dim3(1), dim3(threads), 0,
                            stream, s->flags_d, idx_d, n,
                            (uint32_t)MPIX_FLAG_COMPLETED,
                            (uint32_t)MPIX_FLAG_CLEANUP);

@@ -262,15 +262,26 @@ static void probe_capture_memops(State *s)
         if (hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) !=
             hipSuccess)
             break;
+        /* write magic, then wait for magic+1 (UNSATISFIED): detects both
+         * record failures and no-op replayed waits — on ROCm 7.x the
+         * captured WaitValue32 records fine but does not wait (found the
+         * hard way: gpurun_out/pytest_gpu_fix2.log, stale data, not hang) */
         bool rec_ok =
             hipStreamWriteValue32(st, fd, magic, 0) == hipSuccess &&
-            hipStreamWaitValue32(st, fd, magic, hipStreamWaitValueEq,
+            hipStreamWaitValue32(st, fd, magic + 1, hipStreamWaitValueEq,
                                  0xFFFFFFFFu) == hipSuccess;
         if (hipStreamEndCapture(st, &g) != hipSuccess || !rec_ok || !g)
             break;
         if (hipGraphInstantiate(&ge, g, nullptr, nullptr, 0) != hipSuccess)
             break;
         if (hipGraphLaunch(ge, st) != hipSuccess) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+        if (hipStreamQuery(st) != hipErrorNotReady) {
+            (void)hipGetLastError();
+            break; /* no-op wait: replay passed without its value */
+        }
+        if (s->flags[idx].load() != magic) break; /* write node didn't fire */
+        s->flags[idx].store(magic + 1, std::memory_order_release);
         bool done = false;
         for (int i = 0; i < 2000; i++) {
             hipError_t q = hipStreamQuery(st);
@@ -278,7 +289,11 @@ static void probe_capture_memops(State *s)
             if (q != hipErrorNotReady) break;
             std::this_thread::sleep_for(std::chrono::milliseconds(1));
         }
-        if (!done || s->flags[idx].load() != magic) break;
+        if (!done) {
+            fprintf(stderr, "[mpix] warn: captured WaitValue32 never "
+                    "observed a host store; captures use wait kernels\n");
+            break;
+        }
         ok = true;
     } while (0);
     (void)hipGetLastError();

@@ -373,14 +373,15 @@ int NativeTransport::init()
  * stream→queue mapping (gpurun_out/diag2_*: slots stuck ISSUED,ch_done=0;
  * an attempted dedicated CU-masked queue for the copy stream wedged the
  * runtime outright, diag3).  Deterministic rule instead: use the pull
- * KERNEL only when every wait in the library is a memory operation
- * (probed at init) — then no spin kernel can exist in front of it;
- * otherwise ride SDMA, which compute-queue ordering cannot block. */
+ * KERNEL only while the library has never emitted a spin-wait kernel
+ * (sticky mark_spin_wait in enqueue.cpp — with all memOp tiers probed
+ * functional it never fires); afterwards ride SDMA, which compute-queue
+ * ordering cannot block. */
 static bool pull_kernels_safe()
 {
     State *s = g_state;
-    return s != nullptr && s->use_memops && s->use_graph_memops &&
-           s->use_capture_memops;
+    return s != nullptr && s->use_memops &&
+           !s->spin_wait_kernels.load(std::memory_order_acquire);
 }
 
 void NativeTransport::shutdown()
