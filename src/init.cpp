@@ -480,10 +480,12 @@ extern "C" int MPIX_Init(void)
                                 s->ops[i].ch_done.load(
                                     std::memory_order_relaxed));
                 }
-                fprintf(stderr, " issued=%lu completed=%lu\n",
+                fprintf(stderr, " issued=%lu completed=%lu passes=%lu\n",
                         (unsigned long)s->ops_issued.load(
                             std::memory_order_relaxed),
                         (unsigned long)s->ops_completed.load(
+                            std::memory_order_relaxed),
+                        (unsigned long)s->proxy_passes.load(
                             std::memory_order_relaxed));
             }
         });

@@ -316,6 +316,7 @@ struct State {
     /* stats */
     std::atomic<uint64_t> ops_issued{0};
     std::atomic<uint64_t> ops_completed{0};
+    std::atomic<uint64_t> proxy_passes{0};   /* heartbeat (watchdog) */
     /* issue->complete latency histogram, log2 us buckets 0..19 (>=0.5 ms
      * capped), proxy-thread-only (MPIX_STATS=1) */
     bool stats = false;

@@ -106,6 +106,7 @@ void proxy_main()
 
     while (true) {
         bool did = false;
+        s->proxy_passes.fetch_add(1, std::memory_order_relaxed);
 
         /* adopt newly armed slots */
         int nidx;

@@ -797,8 +797,16 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
         erase_inbound(&m);
         return;
     }
+    if (getenv("MPIX_TRACE") && atoi(getenv("MPIX_TRACE")))
+        fprintf(stderr, "[mpix trace] pull start %lu B via %s (dst=%p src=%p)\n",
+                (unsigned long)n,
+                (n > 0 && n <= copy_kernel_max() && kernel_ok) ? "kernel"
+                                                               : "memcpyAsync",
+                op->buf, src);
     hipEvent_t ev = get_event();
-    (void)hipEventRecord(ev, copy_stream_);
+    hipError_t erec = hipEventRecord(ev, copy_stream_);
+    if (erec != hipSuccess)
+        MPIX_ERR("hipEventRecord(pull) failed: %s", hipGetErrorString(erec));
     copies_.push_back(CopyInflight{op, ev, m.src, m.d.token, st});
     m.dev_copy_started = true;
     erase_inbound(&m);
@@ -813,6 +821,8 @@ int NativeTransport::progress_copies()
             continue;
         }
         if (e != hipSuccess) it->st.err = MPI_ERR_OTHER;
+        if (getenv("MPIX_TRACE") && atoi(getenv("MPIX_TRACE")))
+            fprintf(stderr, "[mpix trace] pull done (e=%d)\n", (int)e);
         put_event(it->ev);
         /* ack the sender, then complete the recv */
         if (ring_has_space(it->src, 1)) {
