@@ -137,6 +137,11 @@ static void probe_graph_memops(State *s)
 {
     s->use_graph_memops = false;
     if (!s->have_gpu || !s->use_memops) return;
+    /* graph batch-memOp nodes ride the same engine as hipStreamBatchMemOp;
+     * if that probe failed, do not even launch this one — a broken wait
+     * node that never completes would poison its hardware queue for the
+     * rest of the process (there is no way to abort a stuck packet) */
+    if (!s->use_batch_memops) return;
     if (env_int("MPIX_DISABLE_GRAPH_MEMOPS", 0)) return;
     const int idx = (int)s->nflags - 1;
     uint32_t *fd = s->flags_d + idx;

@@ -242,6 +242,7 @@ private:
     int dev_;
     bool mpi_mode_;
     bool shut_ = false;
+    bool prio_stream_ = false; /* copy_stream_ owns its hardware queue */
 
     /* helpers */
     InboxView my_inbox(int src) { return inbox_view(seg_[rank_], geom_, src); }
@@ -354,8 +355,31 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
-            hipSuccess) {
+        /* The copy stream must never share a hardware queue with a stream
+         * that can hold a spin-wait kernel: HIP muxes same-priority streams
+         * onto a small HSA queue pool, and GRAPH execution orders a queue
+         * with AQL barrier packets — so a user graph's k_wait_flag node
+         * parked in front of our pull/blit packet blocks it forever, while
+         * the wait itself needs that copy to finish.  Hit deterministically
+         * on the 2nd MPIX_Init generation once stream churn rotated the
+         * mapping (gpurun_out/diag5_loop5.log: 1 KiB memcpyAsync never
+         * completes, proxy alive).  Streams of different PRIORITIES cannot
+         * share a queue, and mpix is the only high-priority stream creator
+         * in the process, so a greatest-priority stream owns its queue.
+         * (A CU-masked stream would too, but wedged this runtime: diag3.) */
+        int lo = 0, hi = 0;
+        if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
+            hi != lo &&
+            hipStreamCreateWithPriority(&copy_stream_, hipStreamNonBlocking,
+                                        hi) == hipSuccess) {
+            prio_stream_ = true; /* dedicated high-priority queue */
+        } else {
+            (void)hipGetLastError();
+            copy_stream_ = nullptr;
+        }
+        if (copy_stream_ == nullptr &&
+            hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
+                hipSuccess) {
             MPIX_ERR("copy stream create failed");
             return -1;
         }
@@ -774,7 +798,12 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * and (b) 16-byte alignment on both sides (user buffers can be
      * arbitrarily offset, e.g. partitioned slices).  Everything else rides
      * hipMemcpyAsync (SDMA / runtime blit). */
-    bool kernel_ok = op->buf_is_device && pull_kernels_safe() &&
+    /* pull kernel allowed when the copy stream owns its hardware queue
+     * (priority stream) — a spinning user graph can then never block it;
+     * on the fallback shared-queue stream only while no spin-wait kernel
+     * has ever been emitted (sticky, see pull_kernels_safe) */
+    bool kernel_ok = op->buf_is_device &&
+                     (prio_stream_ || pull_kernels_safe()) &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
     if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
