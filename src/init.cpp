@@ -232,6 +232,35 @@ static void probe_graph_memops(State *s)
                     "host store; leaving graph memOps off\n");
             break;
         }
+        /* relaunch legs: writes must re-fire and waits must re-block (on
+         * ROCm 7.x replayed memOp nodes can be one-shot no-ops) */
+        s->flags[idx].store(magic, std::memory_order_release);
+        if (hipGraphLaunch(ge, st) != hipSuccess) break;
+        done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        (void)hipGetLastError();
+        if (!done || s->flags[idx].load() != magic + 1) break;
+        s->flags[idx].store(0, std::memory_order_release);
+        if (hipGraphLaunch(ge2, st) != hipSuccess) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+        if (hipStreamQuery(st) != hipErrorNotReady) {
+            (void)hipGetLastError();
+            break; /* relaunched wait was a no-op */
+        }
+        s->flags[idx].store(magic + 2, std::memory_order_release);
+        done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        if (!done) break;
         ok = true;
     } while (0);
     (void)hipGetLastError();
@@ -299,6 +328,28 @@ static void probe_capture_memops(State *s)
                     "observed a host store; captures use wait kernels\n");
             break;
         }
+        /* RELAUNCH must re-fire the write and re-block the wait — on ROCm
+         * 7.x replayed captured memOps are one-shot no-ops (found via
+         * gpurun_out/diag7_cap1.log: iter 0 ok, iter 1 stale), which would
+         * silently break every re-launched captured graph. */
+        if (hipGraphLaunch(ge, st) != hipSuccess) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+        if (hipStreamQuery(st) != hipErrorNotReady) {
+            (void)hipGetLastError();
+            fprintf(stderr, "[mpix] warn: captured memOps do not replay "
+                    "(one-shot); captures use trigger/wait kernels\n");
+            break;
+        }
+        if (s->flags[idx].load() != magic) break; /* write didn't re-fire */
+        s->flags[idx].store(magic + 1, std::memory_order_release);
+        done = false;
+        for (int i = 0; i < 2000; i++) {
+            hipError_t q = hipStreamQuery(st);
+            if (q == hipSuccess) { done = true; break; }
+            if (q != hipErrorNotReady) break;
+            std::this_thread::sleep_for(std::chrono::milliseconds(1));
+        }
+        if (!done) break;
         ok = true;
     } while (0);
     (void)hipGetLastError();
