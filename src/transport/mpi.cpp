@@ -73,7 +73,16 @@ public:
 
     const char *name() const override { return "mpi-passthrough"; }
 
+public:
+    ~MpiTransport() override {
+        for (hipEvent_t ev : evpool_) (void)hipEventDestroy(ev);
+        if (xstream_) (void)hipStreamDestroy(xstream_);
+    }
+
 private:
+    /* staging phases for device-buffer bounces (async on xstream_ so the
+     * proxy never blocks inside a copy — round-1 review finding) */
+    enum Phase { POSTED = 0, STAGE_OUT, STAGE_IN, PRE_POST };
     struct Out {
         Op *op = nullptr;
         MPI_Request req = MPI_REQUEST_NULL;
@@ -81,6 +90,10 @@ private:
         bool bounce_pinned = false;
         bool is_recv = false;
         bool is_part_send = false;
+        int phase = POSTED;
+        hipEvent_t ev = nullptr;       /* staging-copy completion */
+        uint64_t recv_bytes = 0;       /* STAGE_IN: payload size */
+        MPI_Status st{};               /* STAGE_IN: saved recv status */
     };
     static void free_bounce(Out &o) {
         if (!o.bounce) return;
@@ -89,6 +102,42 @@ private:
         o.bounce = nullptr;
     }
     std::list<Out> outstanding_;
+
+    /* staging stream: priority stream = own hardware queue (immune to the
+     * graph spin-kernel barrier hazard, see transport/native.cpp) */
+    hipStream_t xstream_ = nullptr;
+    std::vector<hipEvent_t> evpool_;
+    hipStream_t xstream() {
+        if (xstream_ == nullptr) {
+            int lo = 0, hi = 0;
+            if (hipDeviceGetStreamPriorityRange(&lo, &hi) != hipSuccess ||
+                hi == lo ||
+                hipStreamCreateWithPriority(&xstream_, hipStreamNonBlocking,
+                                            hi) != hipSuccess) {
+                (void)hipGetLastError();
+                (void)hipStreamCreateWithFlags(&xstream_,
+                                               hipStreamNonBlocking);
+            }
+        }
+        return xstream_;
+    }
+    hipEvent_t get_ev() {
+        if (!evpool_.empty()) {
+            hipEvent_t ev = evpool_.back();
+            evpool_.pop_back();
+            return ev;
+        }
+        hipEvent_t ev = nullptr;
+        (void)hipEventCreateWithFlags(&ev, hipEventDisableTiming);
+        return ev;
+    }
+    void put_ev(hipEvent_t ev) { evpool_.push_back(ev); }
+    /* ordering key: MPI non-overtaking is per (comm, dest); a staged send
+     * must not be overtaken by a later send that finished staging first */
+    static uint64_t okey(const Op *op) {
+        return (uint64_t)(uintptr_t)op->comm * 2654435761u ^
+               (uint64_t)(uint32_t)op->peer;
+    }
     std::vector<Op *> parrived_;      /* MPI-4 passthrough recv partitions */
     std::vector<Op *> part_recvs_;    /* emulation: pending PRECV_PART ops */
     struct PartMsg {                  /* emulation: early partition arrivals */
@@ -121,13 +170,30 @@ private:
                 return -1;
             }
             o.bounce_pinned = true;
-            if (!o.is_recv &&
-                hipMemcpy(o.bounce, op->buf, op->bytes, hipMemcpyDeviceToHost)
-                    != hipSuccess) {
-                free_bounce(o);
-                return -1;
+            if (!o.is_recv) {
+                /* async D2H; the MPI_Isend posts from progress() once the
+                 * copy lands (FIFO per (comm,dest) to keep non-overtaking) */
+                if (op->bytes > 0 &&
+                    hipMemcpyAsync(o.bounce, op->buf, op->bytes,
+                                   hipMemcpyDeviceToHost, xstream()) !=
+                        hipSuccess) {
+                    free_bounce(o);
+                    return -1;
+                }
+                o.ev = get_ev();
+                (void)hipEventRecord(o.ev, xstream());
+                o.phase = STAGE_OUT;
+                outstanding_.push_back(o);
+                return 0;
             }
             buf = o.bounce;
+        }
+        if (!o.is_recv && send_pending_before(op)) {
+            /* an earlier send to this (comm,dest) is still staging: posting
+             * now would overtake it — queue for ordered posting instead */
+            o.phase = PRE_POST;
+            outstanding_.push_back(o);
+            return 0;
         }
         int rc;
         if (o.is_recv)
@@ -144,8 +210,83 @@ private:
         return 0;
     }
 
+    bool send_pending_before(const Op *op) const {
+        uint64_t k = okey(op);
+        for (const Out &o : outstanding_)
+            if (!o.is_recv && (o.phase == STAGE_OUT || o.phase == PRE_POST) &&
+                okey(o.op) == k)
+                return true;
+        return false;
+    }
+
+    /* post the MPI send for a staged/queued entry */
+    int post_send(Out &o) {
+        if (o.is_part_send)
+            return MPI_Isend(o.bounce,
+                             (int)(sizeof(PartHdr) + o.op->bytes), MPI_BYTE,
+                             o.op->peer, reserved_part_tag(), o.op->comm,
+                             &o.req);
+        void *buf = o.bounce ? o.bounce : o.op->buf;
+        return MPI_Isend(buf, o.op->count, o.op->datatype, o.op->peer,
+                         o.op->tag, o.op->comm, &o.req);
+    }
+
     void progress_basic() {
+        /* keys whose next send must wait (FIFO posting per (comm,dest)) */
+        std::vector<uint64_t> blocked;
+        auto is_blocked = [&](uint64_t k) {
+            for (uint64_t x : blocked)
+                if (x == k) return true;
+            return false;
+        };
         for (auto it = outstanding_.begin(); it != outstanding_.end();) {
+            if (it->phase == STAGE_OUT || it->phase == PRE_POST) {
+                uint64_t k = okey(it->op);
+                if (is_blocked(k)) {
+                    ++it;
+                    continue;
+                }
+                if (it->ev != nullptr) {
+                    hipError_t e = hipEventQuery(it->ev);
+                    if (e == hipErrorNotReady) {
+                        blocked.push_back(k);
+                        ++it;
+                        continue;
+                    }
+                    put_ev(it->ev);
+                    it->ev = nullptr;
+                    if (e != hipSuccess) {
+                        complete(it->op, -1, it->op->tag, 0, MPI_ERR_OTHER);
+                        free_bounce(*it);
+                        it = outstanding_.erase(it);
+                        continue;
+                    }
+                }
+                if (post_send(*it) != MPI_SUCCESS) {
+                    complete(it->op, -1, it->op->tag, 0, MPI_ERR_OTHER);
+                    free_bounce(*it);
+                    it = outstanding_.erase(it);
+                    continue;
+                }
+                it->phase = POSTED;
+                ++it;
+                continue;
+            }
+            if (it->phase == STAGE_IN) {
+                hipError_t e = hipEventQuery(it->ev);
+                if (e == hipErrorNotReady) {
+                    ++it;
+                    continue;
+                }
+                put_ev(it->ev);
+                it->ev = nullptr;
+                complete(it->op, it->st.MPI_SOURCE, it->st.MPI_TAG,
+                         it->recv_bytes,
+                         e == hipSuccess ? it->st.MPI_ERROR : MPI_ERR_OTHER);
+                free_bounce(*it);
+                it = outstanding_.erase(it);
+                continue;
+            }
             int done = 0;
             MPI_Status st;
             if (MPI_Test(&it->req, &done, &st) != MPI_SUCCESS) {
@@ -166,14 +307,23 @@ private:
                 int cnt = 0, tsz = 0;
                 MPI_Get_count(&st, op->datatype, &cnt);
                 datatype_size(op->datatype, &tsz);
-                if (it->bounce) {
-                    uint64_t n = (uint64_t)cnt * (uint64_t)tsz;
-                    if (n > 0)
-                        (void)hipMemcpy(op->buf, it->bounce, n,
-                                        hipMemcpyHostToDevice);
+                uint64_t n = (uint64_t)cnt * (uint64_t)tsz;
+                if (it->bounce && n > 0) {
+                    /* async H2D; completion in the STAGE_IN phase above */
+                    if (hipMemcpyAsync(op->buf, it->bounce, n,
+                                       hipMemcpyHostToDevice, xstream()) ==
+                        hipSuccess) {
+                        it->ev = get_ev();
+                        (void)hipEventRecord(it->ev, xstream());
+                        it->st = st;
+                        it->recv_bytes = n;
+                        it->phase = STAGE_IN;
+                        ++it;
+                        continue;
+                    }
+                    st.MPI_ERROR = MPI_ERR_OTHER;
                 }
-                complete(op, st.MPI_SOURCE, st.MPI_TAG,
-                         (uint64_t)cnt * (uint64_t)tsz, st.MPI_ERROR);
+                complete(op, st.MPI_SOURCE, st.MPI_TAG, n, st.MPI_ERROR);
             } else {
                 complete(op, -1, op->tag, op->bytes, MPI_SUCCESS);
             }

@@ -229,7 +229,43 @@ private:
     std::vector<uint64_t> in_tail_;  /* consumer-local tail per src */
 
     /* ---- hip ---- */
-    hipStream_t copy_stream_ = nullptr;
+    hipStream_t copy_stream_ = nullptr;      /* current copy stream */
+    hipStream_t copy_stream_plain_ = nullptr; /* pre-switch stream (kept) */
+    bool prio_switched_ = false;
+
+    /* The copy stream must never share a hardware queue with a stream that
+     * holds a SPIN-WAIT kernel: HIP muxes same-priority streams onto a
+     * small HSA queue pool, and GRAPH execution orders its queue with AQL
+     * barrier packets — a k_wait_flag graph node parked in front of our
+     * pull/blit packet blocks it forever, while that wait needs this very
+     * copy to finish (deterministic after stream churn rotated the
+     * mapping: gpurun_out/diag5_loop5.log).  Streams of different
+     * priorities never share a queue, but an always-priority copy stream
+     * costs ~17 us extra half-RTT with two processes on one GPU
+     * (gpurun_out/ci_full2: 54 us vs 37.9), so the switch is LAZY: plain
+     * stream until the library emits its first spin-wait kernel
+     * (mark_spin_wait in enqueue.cpp — ordered before any graph launch
+     * that could contain one), then migrate to a greatest-priority stream.
+     * In-flight copies on the plain stream finish normally (events). */
+    hipStream_t copy_stream() {
+        if (!prio_switched_ && g_state != nullptr &&
+            g_state->spin_wait_kernels.load(std::memory_order_acquire)) {
+            prio_switched_ = true;
+            int lo = 0, hi = 0;
+            hipStream_t ps = nullptr;
+            if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
+                hi != lo &&
+                hipStreamCreateWithPriority(&ps, hipStreamNonBlocking, hi) ==
+                    hipSuccess) {
+                copy_stream_plain_ = copy_stream_;
+                copy_stream_ = ps;
+            } else {
+                (void)hipGetLastError();
+                /* no priorities on this device: nothing safer available */
+            }
+        }
+        return copy_stream_;
+    }
     std::vector<hipEvent_t> event_pool_;
     std::unordered_map<std::string, void *> ipc_open_;   /* handle -> ptr */
     std::unordered_map<const void *, std::pair<void *, hipIpcMemHandle_t>>
@@ -242,7 +278,6 @@ private:
     int dev_;
     bool mpi_mode_;
     bool shut_ = false;
-    bool prio_stream_ = false; /* copy_stream_ owns its hardware queue */
 
     /* helpers */
     InboxView my_inbox(int src) { return inbox_view(seg_[rank_], geom_, src); }
@@ -355,57 +390,13 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        /* The copy stream must never share a hardware queue with a stream
-         * that can hold a spin-wait kernel: HIP muxes same-priority streams
-         * onto a small HSA queue pool, and GRAPH execution orders a queue
-         * with AQL barrier packets — so a user graph's k_wait_flag node
-         * parked in front of our pull/blit packet blocks it forever, while
-         * the wait itself needs that copy to finish.  Hit deterministically
-         * on the 2nd MPIX_Init generation once stream churn rotated the
-         * mapping (gpurun_out/diag5_loop5.log: 1 KiB memcpyAsync never
-         * completes, proxy alive).  Streams of different PRIORITIES cannot
-         * share a queue, and mpix is the only high-priority stream creator
-         * in the process, so a greatest-priority stream owns its queue.
-         * (A CU-masked stream would too, but wedged this runtime: diag3.) */
-        int lo = 0, hi = 0;
-        if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
-            hi != lo &&
-            hipStreamCreateWithPriority(&copy_stream_, hipStreamNonBlocking,
-                                        hi) == hipSuccess) {
-            prio_stream_ = true; /* dedicated high-priority queue */
-        } else {
-            (void)hipGetLastError();
-            copy_stream_ = nullptr;
-        }
-        if (copy_stream_ == nullptr &&
-            hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
-                hipSuccess) {
+        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
+            hipSuccess) {
             MPIX_ERR("copy stream create failed");
             return -1;
         }
     }
     return 0;
-}
-
-/* May the transport run its pull copy as a COMPUTE KERNEL?  HIP muxes
- * streams onto a small HSA hardware-queue pool and serializes a queue with
- * AQL barrier packets during graph execution, so a spin-wait kernel (a
- * k_wait_flag graph node, a captured wait, or the memOps-less stream-wait
- * fallback) that lands on the same queue as the pull-copy kernel blocks it
- * forever — and the spin waits for a flag only that copy can complete.
- * Observed deterministically once enough stream-create cycles rotated the
- * stream→queue mapping (gpurun_out/diag2_*: slots stuck ISSUED,ch_done=0;
- * an attempted dedicated CU-masked queue for the copy stream wedged the
- * runtime outright, diag3).  Deterministic rule instead: use the pull
- * KERNEL only while the library has never emitted a spin-wait kernel
- * (sticky mark_spin_wait in enqueue.cpp — with all memOp tiers probed
- * functional it never fires); afterwards ride SDMA, which compute-queue
- * ordering cannot block. */
-static bool pull_kernels_safe()
-{
-    State *s = g_state;
-    return s != nullptr && s->use_memops &&
-           !s->spin_wait_kernels.load(std::memory_order_acquire);
 }
 
 void NativeTransport::shutdown()
@@ -420,6 +411,8 @@ void NativeTransport::shutdown()
     event_pool_.clear();
     if (copy_stream_) (void)hipStreamDestroy(copy_stream_);
     copy_stream_ = nullptr;
+    if (copy_stream_plain_) (void)hipStreamDestroy(copy_stream_plain_);
+    copy_stream_plain_ = nullptr;
     for (int r = 0; r < (int)seg_.size(); r++)
         if (seg_[r]) munmap(seg_[r], geom_.segment_bytes);
     seg_.clear();
@@ -798,12 +791,10 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * and (b) 16-byte alignment on both sides (user buffers can be
      * arbitrarily offset, e.g. partitioned slices).  Everything else rides
      * hipMemcpyAsync (SDMA / runtime blit). */
-    /* pull kernel allowed when the copy stream owns its hardware queue
-     * (priority stream) — a spinning user graph can then never block it;
-     * on the fallback shared-queue stream only while no spin-wait kernel
-     * has ever been emitted (sticky, see pull_kernels_safe) */
+    /* pull kernel is deadlock-safe in both copy-stream phases: before the
+     * priority switch no spin-wait kernel exists anywhere; after it the
+     * copy stream owns its hardware queue (see copy_stream()) */
     bool kernel_ok = op->buf_is_device &&
-                     (prio_stream_ || pull_kernels_safe()) &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
     if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
@@ -811,10 +802,10 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
         if (blocks == 0) blocks = 1;
         if (blocks > 128) blocks = 128;
         hipLaunchKernelGGL(k_pull_copy, dim3(blocks), dim3(threads), 0,
-                           copy_stream_, op->buf, src, (size_t)n);
+                           copy_stream(), op->buf, src, (size_t)n);
         e = hipGetLastError();
     } else if (n > 0) {
-        e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, copy_stream_);
+        e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, copy_stream());
     }
     if (e != hipSuccess) {
         MPIX_ERR("hipMemcpyAsync(pull %lu B) failed: %s", (unsigned long)n,
@@ -833,7 +824,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
                                                                : "memcpyAsync",
                 op->buf, src);
     hipEvent_t ev = get_event();
-    hipError_t erec = hipEventRecord(ev, copy_stream_);
+    hipError_t erec = hipEventRecord(ev, copy_stream());
     if (erec != hipSuccess)
         MPIX_ERR("hipEventRecord(pull) failed: %s", hipGetErrorString(erec));
     copies_.push_back(CopyInflight{op, ev, m.src, m.d.token, st});
@@ -904,10 +895,10 @@ int NativeTransport::memcpy_auto(void *dst, const void *src, size_t n)
         memcpy(dst, src, n);
         return 0;
     }
-    if (hipMemcpyAsync(dst, src, n, hipMemcpyDefault, copy_stream_) !=
+    if (hipMemcpyAsync(dst, src, n, hipMemcpyDefault, copy_stream()) !=
         hipSuccess)
         return -1;
-    return hipStreamSynchronize(copy_stream_) == hipSuccess ? 0 : -1;
+    return hipStreamSynchronize(copy_stream()) == hipSuccess ? 0 : -1;
 }
 
 /* ----------------------------------------------------------------- factory */
