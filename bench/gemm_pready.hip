@@ -67,7 +67,152 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define MPIX_GEMM_VARIANT 3
 #endif
 
-#if MPIX_GEMM_VARIANT == 3
+#if MPIX_GEMM_VARIANT == 4
+/* Variant 4 — guide §5 "glds + 2 LDS buffers + BK=64" structure on a
+ * 256x256 tile (the measured ~1.1-1.2 PF tier for this shape):
+ *  - 512 threads = 8 waves in a 2(M)x4(N) grid, each owning a 128x64
+ *    C sub-tile = 8x4 mfma_f32_16x16x32_bf16 accumulators;
+ *  - both operands staged HBM->LDS with __builtin_amdgcn_global_load_lds
+ *    (dwordx4), double-buffered (2 x 64 KiB of the 160 KiB LDS);
+ *  - B is taken PRE-TRANSPOSED [N][K] so A and B tiles are both [256][64]
+ *    row-major and every fragment is one 16-B ds_read_b128;
+ *  - glds writes are lane-linear, so the T2 bank-conflict swizzle is
+ *    applied on the per-lane GLOBAL source address (and undone on the
+ *    ds_read): rows alternate 256-B bank-row halves (128-B rows), the
+ *    XOR spreads a 16-lane fragment group over 16 distinct 16-B slots —
+ *    conflict-free.
+ */
+#undef BM
+#undef BN
+#undef BK
+#define BM 256
+#define BN 256
+#define BK 64
+#define V4_THREADS 512
+
+__global__ __launch_bounds__(V4_THREADS)
+void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
+                      bf16 *__restrict__ C, int M, int N, int K,
+                      uint32_t *band_cnt, int nparts, int blocks_per_band,
+                      void *dpreq, int publish)
+{
+    /* one buffer = A[256][64] + B[256][64] bf16 = 64 KiB; two buffers */
+    __shared__ bf16 lds[2 * 2 * BM * BK];
+    auto As = [&](int b) -> bf16 * { return lds + b * 2 * BM * BK; };
+    auto Bs = [&](int b) -> bf16 * { return As(b) + BM * BK; };
+
+    int nwg = gridDim.x;
+    int wg = blockIdx.x;
+    {
+        int q = nwg / 8, r = nwg % 8, xcd = wg % 8;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
+    }
+    int tiles_n = N / BN;
+    int tm = wg / tiles_n, tn = wg % tiles_n;
+
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int wm = wave >> 2, wn = wave & 3; /* 2x4 wave grid */
+    int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+    frag4f acc[8][4] = {};
+    const bf16 *Ab = A + (size_t)tm * BM * K;
+    const bf16 *Bb = Bt + (size_t)tn * BN * K;
+
+    /* glds: each wave stages 4 contiguous 1-KiB LDS chunks per operand
+     * tile (8 waves x 4 passes x 1 KiB = 32 KiB).  LDS dest is
+     * wave-uniform base + lane*16; the per-lane GLOBAL source carries the
+     * swizzle: LDS linear byte o <- global (row = o/128,
+     * col_byte = (o%128) ^ ((row&7)<<4)). */
+    auto glds_tile = [&](const bf16 *gbase, bf16 *lbase, int t) {
+        const char *g0 = (const char *)(gbase) + (size_t)t * BK * sizeof(bf16);
+        #pragma unroll
+        for (int p = 0; p < 4; p++) {
+            unsigned o = ((unsigned)(p * 8 + wave) * 64 + (unsigned)lane) * 16;
+            unsigned row = o >> 7;
+            unsigned cb = (o & 127u) ^ ((row & 7u) << 4);
+            const void *src = g0 + (size_t)row * K * sizeof(bf16) + cb;
+            /* dest uses the lane-linear address; hardware applies
+             * base+lane*16 itself, per-lane addr must match that shape */
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t *)src,
+                (__attribute__((address_space(3))) uint32_t *)
+                    ((char *)lbase + o),
+                16, 0, 0);
+        }
+    };
+    /* fragment read: element (row, k..k+7) lives at row*64 + (k ^ swz) */
+    auto frag_at = [&](const bf16 *base, int row, int kb) -> frag8 {
+        int col = kb ^ ((row & 7) << 3);
+        return *(const frag8 *)&base[row * BK + col];
+    };
+
+    auto compute = [&](int buf) {
+        #pragma unroll
+        for (int kh = 0; kh < 2; kh++) {
+            int kb = kh * 32 + lk8;
+            frag8 af[8], bf[4];
+            #pragma unroll
+            for (int i = 0; i < 8; i++)
+                af[i] = frag_at(As(buf), wm * 128 + i * 16 + lrow, kb);
+            #pragma unroll
+            for (int j = 0; j < 4; j++)
+                bf[j] = frag_at(Bs(buf), wn * 64 + j * 16 + lrow, kb);
+            #pragma unroll
+            for (int i = 0; i < 8; i++)
+                #pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    };
+
+    int tiles = K / BK;
+    glds_tile(Ab, As(0), 0);
+    glds_tile(Bb, Bs(0), 0);
+    __syncthreads(); /* hipcc inserts the vmcnt(0) drain for the glds */
+    for (int t = 0; t < tiles; t++) {
+        int cur = t & 1;
+        if (t + 1 < tiles) {
+            glds_tile(Ab, As(1 - cur), t + 1);
+            glds_tile(Bb, Bs(1 - cur), t + 1);
+        }
+        compute(cur);
+        __syncthreads();
+    }
+
+    size_t crow0 = (size_t)tm * BM + wm * 128;
+    size_t ccol0 = (size_t)tn * BN + wn * 64;
+    #pragma unroll
+    for (int i = 0; i < 8; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            size_t col = ccol0 + j * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; r++) {
+                size_t row = crow0 + i * 16 + (lane >> 4) * 4 + r;
+                C[row * N + col] = (bf16)acc[i][j][r];
+            }
+        }
+
+    if (!publish) return;
+    int band = (int)((size_t)tm * BM * (size_t)nparts / M);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t prev = __hip_atomic_fetch_add(&band_cnt[band], 1,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == (uint32_t)blocks_per_band - 1) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+            MPIX_Pready(band, dpreq);
+        }
+    }
+}
+
+#elif MPIX_GEMM_VARIANT == 3
 /* Variant 3 (round-2 candidate, compile-validated; measure with --check
  * first): BK=64 (half the barriers of v1/v2), double-buffered LDS, and a
  * T14-style register prefetch — tile t+1 streams HBM->VGPRs while tile t
@@ -341,6 +486,12 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
 
 #endif /* MPIX_GEMM_VARIANT */
 
+#if MPIX_GEMM_VARIANT == 4
+#define GEMM_THREADS V4_THREADS
+#else
+#define GEMM_THREADS 256
+#endif
+
 /* ---------------------------------------------------- payload verification
  * Sampled end-to-end check of the RECEIVED buffer in timed mode (not just
  * --check of the GEMM): every VERIFY_EVERY iterations the receiver's Crecv
@@ -445,7 +596,7 @@ int main(int argc, char **argv)
 
     if (check) {
         HIP(hipMemset(band_cnt, 0, nparts * sizeof(uint32_t)));
-        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
+        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(GEMM_THREADS), 0, st,
                            A, B, C, M, N, K, band_cnt, nparts, blocks_per_band,
                            nullptr, 0);
         HIP(hipStreamSynchronize(st));
@@ -455,7 +606,14 @@ int main(int argc, char **argv)
         HIP(hipMemcpy(hC.data(), C, cn * sizeof(bf16), hipMemcpyDeviceToHost));
         std::vector<float> fA(an), fB(bn), fC(cn);
         for (size_t i = 0; i < an; i++) fA[i] = (float)hA[i];
+#if MPIX_GEMM_VARIANT == 4
+        /* kernel takes B pre-transposed [N][K]; reference wants [K][N] */
+        for (int k = 0; k < K; k++)
+            for (int n = 0; n < N; n++)
+                fB[(size_t)k * N + n] = (float)hB[(size_t)n * K + k];
+#else
         for (size_t i = 0; i < bn; i++) fB[i] = (float)hB[i];
+#endif
         host_gemm_ref(fA, fB, fC, M, N, K);
         int bad = 0;
         float worst = 0.f;
@@ -499,12 +657,12 @@ int main(int argc, char **argv)
             CHECK(MPIX_Startall(2, act) == 0);
             HIP(hipMemsetAsync(band_cnt, 0, nparts * sizeof(uint32_t), st));
             if (overlap) {
-                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
+                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(GEMM_THREADS),
                                    0, st, A, B, C, M, N, K, band_cnt, nparts,
                                    blocks_per_band, dps, 1);
                 HIP(hipStreamSynchronize(st));
             } else {
-                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256),
+                hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(GEMM_THREADS),
                                    0, st, A, B, C, M, N, K, band_cnt, nparts,
                                    blocks_per_band, nullptr, 0);
                 HIP(hipStreamSynchronize(st));
@@ -550,7 +708,7 @@ int main(int argc, char **argv)
     /* GEMM-only reference time (no send) for TFLOP/s */
     auto t0 = std::chrono::steady_clock::now();
     for (int i = 0; i < iters; i++) {
-        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(256), 0, st,
+        hipLaunchKernelGGL(gemm_bf16_pready, dim3(grid), dim3(GEMM_THREADS), 0, st,
                            A, B, C, M, N, K, band_cnt, nparts, blocks_per_band,
                            nullptr, 0);
     }
