@@ -67,7 +67,146 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define MPIX_GEMM_VARIANT 3
 #endif
 
-#if MPIX_GEMM_VARIANT == 4
+#if MPIX_GEMM_VARIANT == 5
+/* Variant 5 — v4's 256x256 glds structure pushed to the guide's pipelined
+ * tier: BK=32 tiles in FOUR LDS buffers (4 x 32 KiB), two tiles in flight
+ * across RAW barriers with counted vmcnt (never vmcnt(0) in the loop —
+ * __syncthreads would drain the LDS-DMA), one barrier per K-tile:
+ *
+ *   iter t: s_waitcnt vmcnt(4)      // everything but the newest tile landed
+ *           s_barrier + lgkmcnt(0)  // raw barrier: glds stays in flight
+ *           glds tile t+3 -> buf[(t+3)&3]   // overwrite of buf[(t-1)&3]
+ *                                           // is ordered by the barrier
+ *           compute tile t from buf[t&3]
+ *
+ * 64-B LDS rows (BK=32): rows r, r+4, r+8, r+12 share a 256-B bank row,
+ * so the source-side swizzle XORs the 16-B slot with (row>>2)&3 —
+ * fragment groups hit 16 distinct (quarter, slot) pairs, conflict-free.
+ */
+#undef BM
+#undef BN
+#undef BK
+#define BM 256
+#define BN 256
+#define BK 32
+#define V4_THREADS 512
+
+__global__ __launch_bounds__(V4_THREADS)
+void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
+                      bf16 *__restrict__ C, int M, int N, int K,
+                      uint32_t *band_cnt, int nparts, int blocks_per_band,
+                      void *dpreq, int publish)
+{
+    __shared__ bf16 lds[4 * 2 * BM * BK]; /* 4 bufs x (A+B) x 16 KiB */
+    auto As = [&](int b) -> bf16 * { return lds + b * 2 * BM * BK; };
+    auto Bs = [&](int b) -> bf16 * { return As(b) + BM * BK; };
+
+    int nwg = gridDim.x;
+    int wg = blockIdx.x;
+    {
+        int q = nwg / 8, r = nwg % 8, xcd = wg % 8;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
+    }
+    int tiles_n = N / BN;
+    int tm = wg / tiles_n, tn = wg % tiles_n;
+
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int wm = wave >> 2, wn = wave & 3;
+    int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+    frag4f acc[8][4] = {};
+    const bf16 *Ab = A + (size_t)tm * BM * K;
+    const bf16 *Bb = Bt + (size_t)tn * BN * K;
+
+    /* 16-KiB tile = 2 glds passes x 8 KiB; swizzle on the global source */
+    auto glds_tile = [&](const bf16 *gbase, bf16 *lbase, int t) {
+        const char *g0 = (const char *)(gbase) + (size_t)t * BK * sizeof(bf16);
+        #pragma unroll
+        for (int p = 0; p < 2; p++) {
+            unsigned o = ((unsigned)(p * 8 + wave) * 64 + (unsigned)lane) * 16;
+            unsigned row = o >> 6;
+            unsigned cb = (o & 63u) ^ (((row >> 2) & 3u) << 4);
+            const void *src = g0 + (size_t)row * K * sizeof(bf16) + cb;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t *)src,
+                (__attribute__((address_space(3))) uint32_t *)
+                    ((char *)lbase + o),
+                16, 0, 0);
+        }
+    };
+    auto frag_at = [&](const bf16 *base, int row, int kb) -> frag8 {
+        int col = kb ^ (((row >> 2) & 3) << 3);
+        return *(const frag8 *)&base[row * BK + col];
+    };
+
+    auto compute = [&](int buf) {
+        frag8 af[8], bf[4];
+        #pragma unroll
+        for (int i = 0; i < 8; i++)
+            af[i] = frag_at(As(buf), wm * 128 + i * 16 + lrow, lk8);
+        #pragma unroll
+        for (int j = 0; j < 4; j++)
+            bf[j] = frag_at(Bs(buf), wn * 64 + j * 16 + lrow, lk8);
+        #pragma unroll
+        for (int i = 0; i < 8; i++)
+            #pragma unroll
+            for (int j = 0; j < 4; j++)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[i], bf[j], acc[i][j], 0, 0, 0);
+    };
+
+    int tiles = K / BK;
+    glds_tile(Ab, As(0), 0);
+    glds_tile(Bb, Bs(0), 0);
+    if (tiles > 1) { glds_tile(Ab, As(1), 1); glds_tile(Bb, Bs(1), 1); }
+    if (tiles > 2) { glds_tile(Ab, As(2), 2); glds_tile(Bb, Bs(2), 2); }
+    for (int t = 0; t < tiles; t++) {
+        if (t + 2 < tiles)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        if (t + 3 < tiles) {
+            glds_tile(Ab, As((t + 3) & 3), t + 3);
+            glds_tile(Bb, Bs((t + 3) & 3), t + 3);
+        }
+        compute(t & 3);
+    }
+
+    size_t crow0 = (size_t)tm * BM + wm * 128;
+    size_t ccol0 = (size_t)tn * BN + wn * 64;
+    #pragma unroll
+    for (int i = 0; i < 8; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            size_t col = ccol0 + j * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; r++) {
+                size_t row = crow0 + i * 16 + (lane >> 4) * 4 + r;
+                C[row * N + col] = (bf16)acc[i][j][r];
+            }
+        }
+
+    if (!publish) return;
+    int band = (int)((size_t)tm * BM * (size_t)nparts / M);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t prev = __hip_atomic_fetch_add(&band_cnt[band], 1,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == (uint32_t)blocks_per_band - 1) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+            MPIX_Pready(band, dpreq);
+        }
+    }
+}
+
+#elif MPIX_GEMM_VARIANT == 4
 /* Variant 4 — guide §5 "glds + 2 LDS buffers + BK=64" structure on a
  * 256x256 tile (the measured ~1.1-1.2 PF tier for this shape):
  *  - 512 threads = 8 waves in a 2(M)x4(N) grid, each owning a 128x64
@@ -486,7 +625,7 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ B,
 
 #endif /* MPIX_GEMM_VARIANT */
 
-#if MPIX_GEMM_VARIANT == 4
+#if MPIX_GEMM_VARIANT >= 4
 #define GEMM_THREADS V4_THREADS
 #else
 #define GEMM_THREADS 256
@@ -606,7 +745,7 @@ int main(int argc, char **argv)
         HIP(hipMemcpy(hC.data(), C, cn * sizeof(bf16), hipMemcpyDeviceToHost));
         std::vector<float> fA(an), fB(bn), fC(cn);
         for (size_t i = 0; i < an; i++) fA[i] = (float)hA[i];
-#if MPIX_GEMM_VARIANT == 4
+#if MPIX_GEMM_VARIANT >= 4
         /* kernel takes B pre-transposed [N][K]; reference wants [K][N] */
         for (int k = 0; k < K; k++)
             for (int n = 0; n < N; n++)
