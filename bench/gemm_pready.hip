@@ -62,9 +62,11 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define BN 128
 #define BK 32
 #ifndef MPIX_GEMM_VARIANT
-/* default = variant 3: BK=64 double-buffered LDS + register prefetch,
- * 510.9 TF @8192^3 vs 476.1 (v2) / 236 (v1) — profiles/r02_gemm_v3_8k.json */
-#define MPIX_GEMM_VARIANT 3
+/* default = variant 4 (256^2 tile, BK=64, 8-wave glds double-buffer,
+ * source-side bank swizzle): 909.8 TF @8192^3 — ladder: v1 236, v2 476,
+ * v3 511, v5 (BK=32 4-buf raw-barrier pipeline) 722 — see
+ * profiles/r02_gemm_ladder.md */
+#define MPIX_GEMM_VARIANT 4
 #endif
 
 #if MPIX_GEMM_VARIANT == 5
