@@ -229,8 +229,14 @@ private:
     std::vector<uint64_t> in_tail_;  /* consumer-local tail per src */
 
     /* ---- hip ---- */
-    hipStream_t copy_stream_ = nullptr;      /* current copy stream */
-    hipStream_t copy_stream_plain_ = nullptr; /* pre-switch stream (kept) */
+    /* Copy-stream POOL: concurrent partition pulls overlap instead of
+     * serializing behind one stream's launch+transfer latency (the 64 x
+     * 4 MiB Psend step was ~9 us per partition end-to-end on one stream —
+     * latency-bound, not bandwidth-bound). */
+    static constexpr int N_COPY_STREAMS = 4;
+    hipStream_t copy_streams_[N_COPY_STREAMS] = {};
+    hipStream_t copy_plain_[N_COPY_STREAMS] = {}; /* pre-switch set (kept) */
+    unsigned copy_rr_ = 0;
     bool prio_switched_ = false;
 
     /* The copy stream must never share a hardware queue with a stream that
@@ -247,24 +253,38 @@ private:
      * (mark_spin_wait in enqueue.cpp — ordered before any graph launch
      * that could contain one), then migrate to a greatest-priority stream.
      * In-flight copies on the plain stream finish normally (events). */
-    hipStream_t copy_stream() {
-        if (!prio_switched_ && g_state != nullptr &&
-            g_state->spin_wait_kernels.load(std::memory_order_acquire)) {
-            prio_switched_ = true;
-            int lo = 0, hi = 0;
-            hipStream_t ps = nullptr;
-            if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
-                hi != lo &&
-                hipStreamCreateWithPriority(&ps, hipStreamNonBlocking, hi) ==
-                    hipSuccess) {
-                copy_stream_plain_ = copy_stream_;
-                copy_stream_ = ps;
-            } else {
-                (void)hipGetLastError();
-                /* no priorities on this device: nothing safer available */
-            }
+    void maybe_switch_prio() {
+        if (prio_switched_ || g_state == nullptr ||
+            !g_state->spin_wait_kernels.load(std::memory_order_acquire))
+            return;
+        prio_switched_ = true;
+        int lo = 0, hi = 0;
+        if (hipDeviceGetStreamPriorityRange(&lo, &hi) != hipSuccess ||
+            hi == lo) {
+            (void)hipGetLastError();
+            return; /* no priorities on this device: nothing safer available */
         }
-        return copy_stream_;
+        for (int i = 0; i < N_COPY_STREAMS; i++) {
+            hipStream_t ps = nullptr;
+            if (hipStreamCreateWithPriority(&ps, hipStreamNonBlocking, hi) !=
+                hipSuccess) {
+                (void)hipGetLastError();
+                break;
+            }
+            copy_plain_[i] = copy_streams_[i];
+            copy_streams_[i] = ps;
+        }
+    }
+    /* round-robin stream for an independent copy (per-message ordering is
+     * by the event recorded on the same stream as the copy) */
+    hipStream_t copy_stream() {
+        maybe_switch_prio();
+        return copy_streams_[copy_rr_++ % N_COPY_STREAMS];
+    }
+    /* fixed stream for synchronous staging (memcpy_auto) */
+    hipStream_t copy_stream0() {
+        maybe_switch_prio();
+        return copy_streams_[0];
     }
     std::vector<hipEvent_t> event_pool_;
     std::unordered_map<std::string, void *> ipc_open_;   /* handle -> ptr */
@@ -390,10 +410,12 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        if (hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking) !=
-            hipSuccess) {
-            MPIX_ERR("copy stream create failed");
-            return -1;
+        for (int i = 0; i < N_COPY_STREAMS; i++) {
+            if (hipStreamCreateWithFlags(&copy_streams_[i],
+                                         hipStreamNonBlocking) != hipSuccess) {
+                MPIX_ERR("copy stream create failed");
+                return -1;
+            }
         }
     }
     return 0;
@@ -409,10 +431,12 @@ void NativeTransport::shutdown()
     ipc_open_.clear();
     for (hipEvent_t ev : event_pool_) (void)hipEventDestroy(ev);
     event_pool_.clear();
-    if (copy_stream_) (void)hipStreamDestroy(copy_stream_);
-    copy_stream_ = nullptr;
-    if (copy_stream_plain_) (void)hipStreamDestroy(copy_stream_plain_);
-    copy_stream_plain_ = nullptr;
+    for (int i = 0; i < N_COPY_STREAMS; i++) {
+        if (copy_streams_[i]) (void)hipStreamDestroy(copy_streams_[i]);
+        copy_streams_[i] = nullptr;
+        if (copy_plain_[i]) (void)hipStreamDestroy(copy_plain_[i]);
+        copy_plain_[i] = nullptr;
+    }
     for (int r = 0; r < (int)seg_.size(); r++)
         if (seg_[r]) munmap(seg_[r], geom_.segment_bytes);
     seg_.clear();
@@ -796,16 +820,17 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * copy stream owns its hardware queue (see copy_stream()) */
     bool kernel_ok = op->buf_is_device &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
+    hipStream_t cs = copy_stream();
     if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
         unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
         if (blocks == 0) blocks = 1;
-        if (blocks > 128) blocks = 128;
+        if (blocks > 256) blocks = 256;
         hipLaunchKernelGGL(k_pull_copy, dim3(blocks), dim3(threads), 0,
-                           copy_stream(), op->buf, src, (size_t)n);
+                           cs, op->buf, src, (size_t)n);
         e = hipGetLastError();
     } else if (n > 0) {
-        e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, copy_stream());
+        e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, cs);
     }
     if (e != hipSuccess) {
         MPIX_ERR("hipMemcpyAsync(pull %lu B) failed: %s", (unsigned long)n,
@@ -824,7 +849,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
                                                                : "memcpyAsync",
                 op->buf, src);
     hipEvent_t ev = get_event();
-    hipError_t erec = hipEventRecord(ev, copy_stream());
+    hipError_t erec = hipEventRecord(ev, cs);
     if (erec != hipSuccess)
         MPIX_ERR("hipEventRecord(pull) failed: %s", hipGetErrorString(erec));
     copies_.push_back(CopyInflight{op, ev, m.src, m.d.token, st});
@@ -895,10 +920,10 @@ int NativeTransport::memcpy_auto(void *dst, const void *src, size_t n)
         memcpy(dst, src, n);
         return 0;
     }
-    if (hipMemcpyAsync(dst, src, n, hipMemcpyDefault, copy_stream()) !=
-        hipSuccess)
+    hipStream_t cs = copy_stream0();
+    if (hipMemcpyAsync(dst, src, n, hipMemcpyDefault, cs) != hipSuccess)
         return -1;
-    return hipStreamSynchronize(copy_stream()) == hipSuccess ? 0 : -1;
+    return hipStreamSynchronize(cs) == hipSuccess ? 0 : -1;
 }
 
 /* ----------------------------------------------------------------- factory */
