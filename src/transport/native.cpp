@@ -229,15 +229,32 @@ private:
     std::vector<uint64_t> in_tail_;  /* consumer-local tail per src */
 
     /* ---- hip ---- */
-    /* Copy-stream POOL: concurrent partition pulls overlap instead of
-     * serializing behind one stream's launch+transfer latency (the 64 x
-     * 4 MiB Psend step was ~9 us per partition end-to-end on one stream —
-     * latency-bound, not bandwidth-bound). */
+    /* Copy-stream POOL.  Stream 0 is PLAIN and carries small pulls plus
+     * synchronous staging (lowest latency: a priority queue costs ~17 us
+     * extra half-RTT, ci_full2); streams 1..3 are created at GREATEST
+     * PRIORITY and carry large pulls concurrently — large partitioned
+     * transfers were latency-bound at ~9 us/partition serialized on one
+     * stream.  The high-priority set can never share a hardware queue
+     * with user streams (different priority = different queue), which
+     * matters beyond spin kernels: an unsatisfied hipStreamWaitValue32 is
+     * a queue-parking packet, so a SAME-priority copy stream that aliases
+     * a user stream's queue deadlocks exactly like the graph spin-kernel
+     * case (observed when a 4-plain-stream pool pushed total streams past
+     * the 4-queue pool: every transfer hung).  Stream 0 carries the same
+     * small residual aliasing risk as the original single-stream design
+     * (mitigated by the lazy priority upgrade below). */
     static constexpr int N_COPY_STREAMS = 4;
     hipStream_t copy_streams_[N_COPY_STREAMS] = {};
-    hipStream_t copy_plain_[N_COPY_STREAMS] = {}; /* pre-switch set (kept) */
+    hipStream_t copy_plain0_ = nullptr; /* pre-upgrade stream 0 (kept) */
     unsigned copy_rr_ = 0;
     bool prio_switched_ = false;
+    static uint64_t pool_min_bytes() {
+        static const uint64_t v = [] {
+            const char *e = getenv("MPIX_COPY_POOL_MIN");
+            return e ? (uint64_t)atoll(e) : (uint64_t)(256 << 10);
+        }();
+        return v;
+    }
 
     /* The copy stream must never share a hardware queue with a stream that
      * holds a SPIN-WAIT kernel: HIP muxes same-priority streams onto a
@@ -259,27 +276,26 @@ private:
             return;
         prio_switched_ = true;
         int lo = 0, hi = 0;
-        if (hipDeviceGetStreamPriorityRange(&lo, &hi) != hipSuccess ||
-            hi == lo) {
-            (void)hipGetLastError();
-            return; /* no priorities on this device: nothing safer available */
-        }
-        for (int i = 0; i < N_COPY_STREAMS; i++) {
-            hipStream_t ps = nullptr;
-            if (hipStreamCreateWithPriority(&ps, hipStreamNonBlocking, hi) !=
+        hipStream_t ps = nullptr;
+        if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
+            hi != lo &&
+            hipStreamCreateWithPriority(&ps, hipStreamNonBlocking, hi) ==
                 hipSuccess) {
-                (void)hipGetLastError();
-                break;
-            }
-            copy_plain_[i] = copy_streams_[i];
-            copy_streams_[i] = ps;
+            copy_plain0_ = copy_streams_[0];
+            copy_streams_[0] = ps;
+        } else {
+            (void)hipGetLastError();
         }
     }
-    /* round-robin stream for an independent copy (per-message ordering is
-     * by the event recorded on the same stream as the copy) */
-    hipStream_t copy_stream() {
+    /* stream for an independent pull: small payloads ride the plain
+     * stream 0 (latency), large ones rotate over the priority set 1..3
+     * (bandwidth; per-message ordering is by the event recorded on the
+     * same stream as the copy) */
+    hipStream_t copy_stream(uint64_t bytes) {
         maybe_switch_prio();
-        return copy_streams_[copy_rr_++ % N_COPY_STREAMS];
+        if (bytes <= pool_min_bytes() || copy_streams_[1] == nullptr)
+            return copy_streams_[0];
+        return copy_streams_[1 + copy_rr_++ % (N_COPY_STREAMS - 1)];
     }
     /* fixed stream for synchronous staging (memcpy_auto) */
     hipStream_t copy_stream0() {
@@ -410,12 +426,25 @@ int NativeTransport::init()
     in_tail_.assign(size_, 0);
 
     if (have_gpu_) {
-        for (int i = 0; i < N_COPY_STREAMS; i++) {
-            if (hipStreamCreateWithFlags(&copy_streams_[i],
-                                         hipStreamNonBlocking) != hipSuccess) {
-                MPIX_ERR("copy stream create failed");
-                return -1;
+        if (hipStreamCreateWithFlags(&copy_streams_[0],
+                                     hipStreamNonBlocking) != hipSuccess) {
+            MPIX_ERR("copy stream create failed");
+            return -1;
+        }
+        int lo = 0, hi = 0;
+        if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
+            hi != lo) {
+            for (int i = 1; i < N_COPY_STREAMS; i++) {
+                if (hipStreamCreateWithPriority(&copy_streams_[i],
+                                                hipStreamNonBlocking, hi) !=
+                    hipSuccess) {
+                    (void)hipGetLastError();
+                    copy_streams_[i] = nullptr;
+                    break;
+                }
             }
+        } else {
+            (void)hipGetLastError();
         }
     }
     return 0;
@@ -434,9 +463,9 @@ void NativeTransport::shutdown()
     for (int i = 0; i < N_COPY_STREAMS; i++) {
         if (copy_streams_[i]) (void)hipStreamDestroy(copy_streams_[i]);
         copy_streams_[i] = nullptr;
-        if (copy_plain_[i]) (void)hipStreamDestroy(copy_plain_[i]);
-        copy_plain_[i] = nullptr;
     }
+    if (copy_plain0_) (void)hipStreamDestroy(copy_plain0_);
+    copy_plain0_ = nullptr;
     for (int r = 0; r < (int)seg_.size(); r++)
         if (seg_[r]) munmap(seg_[r], geom_.segment_bytes);
     seg_.clear();
@@ -820,7 +849,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * copy stream owns its hardware queue (see copy_stream()) */
     bool kernel_ok = op->buf_is_device &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
-    hipStream_t cs = copy_stream();
+    hipStream_t cs = copy_stream(n);
     if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
         unsigned threads = 256;
         unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
