@@ -46,24 +46,33 @@
 
 namespace mpix {
 
-/* Peer-pull copy kernel for small/medium device payloads: a shader copy
- * over the peer mapping dodges the SDMA engine's ~40 us submit-to-complete
- * latency (measured: the 32 KiB..2 MiB pingpong points sat flat at ~80 us
- * half-RTT on SDMA vs ~42 us via the runtime's blit path).  Large copies
- * stay on hipMemcpyAsync/SDMA, which frees CUs and wins on bandwidth.
- * 16-byte vector copies with a scalar tail; grid sized for the payload. */
-__global__ void k_pull_copy(void *__restrict__ dst, const void *__restrict__ src,
-                            size_t n)
+/* Batched pull: one launch copies every partition/message that became
+ * ready in the same progress pass (a 64 x 4 MiB partitioned step was
+ * latency-bound at ~9 us per serialized launch+event).  All blocks stripe
+ * over each copy in turn — balanced regardless of size mix.  Args live in
+ * pinned mapped memory (an 8-slot ring reused after the batch's event). */
+#define MPIX_PULL_BATCH 64
+struct PullArg {
+    void *dst;
+    const void *src;
+    size_t n16;   /* 16-byte vectors */
+    size_t tail;  /* leftover bytes after n16*16 */
+};
+
+__global__ void k_pull_multi(const PullArg *__restrict__ args, int ncopies)
 {
-    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    size_t nv = n / 16;
-    const uint4 *s4 = (const uint4 *)src;
-    uint4 *d4 = (uint4 *)dst;
-    for (size_t v = i; v < nv; v += (size_t)gridDim.x * blockDim.x)
-        d4[v] = s4[v];
-    if (i == 0)
-        for (size_t b = nv * 16; b < n; b++)
-            ((char *)dst)[b] = ((const char *)src)[b];
+    size_t gid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (int c = 0; c < ncopies; c++) {
+        const uint4 *s4 = (const uint4 *)args[c].src;
+        uint4 *d4 = (uint4 *)args[c].dst;
+        size_t nv = args[c].n16;
+        for (size_t v = gid; v < nv; v += stride) d4[v] = s4[v];
+        if (gid == 0)
+            for (size_t b = 0; b < args[c].tail; b++)
+                ((char *)args[c].dst)[nv * 16 + b] =
+                    ((const char *)args[c].src)[nv * 16 + b];
+    }
 }
 
 /* Sender-push threshold for small DEVICE payloads: stage D2H into the shm
@@ -75,15 +84,6 @@ static uint64_t dev_push_max()
     static const uint64_t v = [] {
         const char *e = getenv("MPIX_DEV_PUSH_MAX");
         return e ? (uint64_t)atoll(e) : (uint64_t)0;
-    }();
-    return v;
-}
-
-static uint64_t copy_kernel_max()
-{
-    static const uint64_t v = [] {
-        const char *e = getenv("MPIX_COPY_KERNEL_MAX");
-        return e ? (uint64_t)atoll(e) : (uint64_t)(2 << 20);
     }();
     return v;
 }
@@ -220,6 +220,29 @@ private:
     };
     std::list<CopyInflight> copies_;
 
+    /* pulls collected this progress pass, flushed as ONE kernel launch */
+    struct PendingPull {
+        Op *op;
+        int src;
+        uint64_t token;
+        ChStatus st;
+        void *dst;
+        const void *csrc;
+        uint64_t n;
+    };
+    std::vector<PendingPull> pend_pulls_;
+    struct PullBatch {
+        hipEvent_t ev;
+        std::vector<PendingPull> items;
+    };
+    std::list<PullBatch> batches_;
+    /* pinned arg ring: slot i free when no live batch references it */
+    static constexpr int ARG_SLOTS = 8;
+    PullArg *args_h_ = nullptr;   /* pinned, ARG_SLOTS * MPIX_PULL_BATCH */
+    PullArg *args_d_ = nullptr;
+    int arg_slot_ = 0;
+    int arg_slots_live_ = 0;
+
     /* ---- shm ---- */
     ShmGeom geom_{};
     std::string my_seg_name_;
@@ -246,7 +269,6 @@ private:
     static constexpr int N_COPY_STREAMS = 4;
     hipStream_t copy_streams_[N_COPY_STREAMS] = {};
     hipStream_t copy_plain0_ = nullptr; /* pre-upgrade stream 0 (kept) */
-    unsigned copy_rr_ = 0;
     bool prio_switched_ = false;
     static uint64_t pool_min_bytes() {
         static const uint64_t v = [] {
@@ -295,7 +317,7 @@ private:
         maybe_switch_prio();
         if (bytes <= pool_min_bytes() || copy_streams_[1] == nullptr)
             return copy_streams_[0];
-        return copy_streams_[1 + copy_rr_++ % (N_COPY_STREAMS - 1)];
+        return copy_streams_[1];
     }
     /* fixed stream for synchronous staging (memcpy_auto) */
     hipStream_t copy_stream0() {
@@ -343,6 +365,7 @@ private:
     int progress_sends();
     int drain_inbox(int src);
     int progress_copies();
+    void flush_pulls();
     void handle_desc(int src, const Desc &d, const char *stage_base);
     void try_match_new_inbound(InboundMsg &m);
     void attach(InboundMsg &m, Op *op);
@@ -431,20 +454,30 @@ int NativeTransport::init()
             MPIX_ERR("copy stream create failed");
             return -1;
         }
+        /* ONE extra priority stream only: every active hardware queue
+         * beyond ~4 per process oversubscribes the HW scheduler and adds
+         * tens of us to every wait (measured: a 4-plain pool pushed the
+         * 8-B half-RTT from 36.5 to 50 us even for traffic on stream 0) */
         int lo = 0, hi = 0;
         if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
             hi != lo) {
-            for (int i = 1; i < N_COPY_STREAMS; i++) {
-                if (hipStreamCreateWithPriority(&copy_streams_[i],
-                                                hipStreamNonBlocking, hi) !=
-                    hipSuccess) {
-                    (void)hipGetLastError();
-                    copy_streams_[i] = nullptr;
-                    break;
-                }
+            if (hipStreamCreateWithPriority(&copy_streams_[1],
+                                            hipStreamNonBlocking, hi) !=
+                hipSuccess) {
+                (void)hipGetLastError();
+                copy_streams_[1] = nullptr;
             }
         } else {
             (void)hipGetLastError();
+        }
+        if (hipHostMalloc((void **)&args_h_,
+                          (size_t)ARG_SLOTS * MPIX_PULL_BATCH *
+                              sizeof(PullArg),
+                          hipHostMallocMapped) != hipSuccess ||
+            hipHostGetDevicePointer((void **)&args_d_, args_h_, 0) !=
+                hipSuccess) {
+            MPIX_ERR("pull-arg ring alloc failed");
+            return -1;
         }
     }
     return 0;
@@ -466,6 +499,8 @@ void NativeTransport::shutdown()
     }
     if (copy_plain0_) (void)hipStreamDestroy(copy_plain0_);
     copy_plain0_ = nullptr;
+    if (args_h_) (void)hipHostFree(args_h_);
+    args_h_ = nullptr;
     for (int r = 0; r < (int)seg_.size(); r++)
         if (seg_[r]) munmap(seg_[r], geom_.segment_bytes);
     seg_.clear();
@@ -535,6 +570,7 @@ void NativeTransport::progress()
 {
     progress_sends();
     for (int s = 0; s < size_; s++) drain_inbox(s);
+    flush_pulls();
     progress_copies();
 
     /* retry queued DONE acks */
@@ -849,16 +885,17 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
      * copy stream owns its hardware queue (see copy_stream()) */
     bool kernel_ok = op->buf_is_device &&
                      ((((uintptr_t)op->buf) | ((uintptr_t)src)) & 15) == 0;
+    if (n > 0 && kernel_ok) {
+        /* defer: every pull that matched in this progress pass goes out in
+         * ONE k_pull_multi launch (flush_pulls) */
+        pend_pulls_.push_back(PendingPull{op, m.src, m.d.token, st,
+                                          op->buf, src, n});
+        m.dev_copy_started = true;
+        erase_inbound(&m);
+        return;
+    }
     hipStream_t cs = copy_stream(n);
-    if (n > 0 && n <= copy_kernel_max() && kernel_ok) {
-        unsigned threads = 256;
-        unsigned blocks = (unsigned)((n / 16 + threads - 1) / threads);
-        if (blocks == 0) blocks = 1;
-        if (blocks > 256) blocks = 256;
-        hipLaunchKernelGGL(k_pull_copy, dim3(blocks), dim3(threads), 0,
-                           cs, op->buf, src, (size_t)n);
-        e = hipGetLastError();
-    } else if (n > 0) {
+    if (n > 0) {
         e = hipMemcpyAsync(op->buf, src, n, hipMemcpyDefault, cs);
     }
     if (e != hipSuccess) {
@@ -872,11 +909,8 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
         return;
     }
     if (getenv("MPIX_TRACE") && atoi(getenv("MPIX_TRACE")))
-        fprintf(stderr, "[mpix trace] pull start %lu B via %s (dst=%p src=%p)\n",
-                (unsigned long)n,
-                (n > 0 && n <= copy_kernel_max() && kernel_ok) ? "kernel"
-                                                               : "memcpyAsync",
-                op->buf, src);
+        fprintf(stderr, "[mpix trace] pull start %lu B via memcpyAsync "
+                "(dst=%p src=%p)\n", (unsigned long)n, op->buf, src);
     hipEvent_t ev = get_event();
     hipError_t erec = hipEventRecord(ev, cs);
     if (erec != hipSuccess)
@@ -886,8 +920,83 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
     erase_inbound(&m);
 }
 
+void NativeTransport::flush_pulls()
+{
+    while (!pend_pulls_.empty()) {
+        if (arg_slots_live_ >= ARG_SLOTS) return; /* ring full: next pass */
+        int n = (int)pend_pulls_.size();
+        if (n > MPIX_PULL_BATCH) n = MPIX_PULL_BATCH;
+        PullArg *slot_h = args_h_ + (size_t)arg_slot_ * MPIX_PULL_BATCH;
+        PullArg *slot_d = args_d_ + (size_t)arg_slot_ * MPIX_PULL_BATCH;
+        uint64_t total = 0;
+        for (int i = 0; i < n; i++) {
+            const PendingPull &pp = pend_pulls_[i];
+            slot_h[i].dst = pp.dst;
+            slot_h[i].src = pp.csrc;
+            slot_h[i].n16 = pp.n / 16;
+            slot_h[i].tail = pp.n % 16;
+            total += pp.n;
+        }
+        hipStream_t cs = copy_stream(total);
+        unsigned threads = 256;
+        unsigned blocks = (unsigned)((total / 16 + threads - 1) / threads);
+        if (blocks == 0) blocks = 1;
+        if (blocks > 512) blocks = 512;
+        hipLaunchKernelGGL(k_pull_multi, dim3(blocks), dim3(threads), 0, cs,
+                           slot_d, n);
+        hipError_t e = hipGetLastError();
+        if (e != hipSuccess) {
+            MPIX_ERR("k_pull_multi launch failed: %s", hipGetErrorString(e));
+            for (int i = 0; i < n; i++) {
+                PendingPull &pp = pend_pulls_[i];
+                pp.st.err = MPI_ERR_OTHER;
+                pp.op->ch_status = pp.st;
+                pp.op->ch_done.store(1, std::memory_order_release);
+                pending_done_.emplace_back(pp.src, pp.token);
+            }
+            pend_pulls_.erase(pend_pulls_.begin(), pend_pulls_.begin() + n);
+            continue;
+        }
+        PullBatch b;
+        b.ev = get_event();
+        (void)hipEventRecord(b.ev, cs);
+        b.items.assign(pend_pulls_.begin(), pend_pulls_.begin() + n);
+        pend_pulls_.erase(pend_pulls_.begin(), pend_pulls_.begin() + n);
+        batches_.push_back(std::move(b));
+        arg_slot_ = (arg_slot_ + 1) % ARG_SLOTS;
+        arg_slots_live_++;
+        if (getenv("MPIX_TRACE") && atoi(getenv("MPIX_TRACE")))
+            fprintf(stderr, "[mpix trace] pull batch n=%d total=%lu B\n", n,
+                    (unsigned long)total);
+    }
+}
+
 int NativeTransport::progress_copies()
 {
+    for (auto it = batches_.begin(); it != batches_.end();) {
+        hipError_t e = hipEventQuery(it->ev);
+        if (e == hipErrorNotReady) {
+            ++it;
+            continue;
+        }
+        put_event(it->ev);
+        for (PendingPull &pp : it->items) {
+            if (e != hipSuccess) pp.st.err = MPI_ERR_OTHER;
+            if (ring_has_space(pp.src, 1)) {
+                Desc d;
+                d.type = DESC_DONE;
+                d.token = pp.token;
+                d.src_world = rank_;
+                emit_desc(pp.src, d);
+            } else {
+                pending_done_.emplace_back(pp.src, pp.token);
+            }
+            pp.op->ch_status = pp.st;
+            pp.op->ch_done.store(1, std::memory_order_release);
+        }
+        arg_slots_live_--;
+        it = batches_.erase(it);
+    }
     for (auto it = copies_.begin(); it != copies_.end();) {
         hipError_t e = hipEventQuery(it->ev);
         if (e == hipErrorNotReady) {
