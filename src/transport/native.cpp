@@ -79,6 +79,12 @@ __global__ void k_pull_multi(const PullArg *__restrict__ args, int ncopies)
  * chunk ring like a host message (receiver H2D's it out), skipping the
  * NOTIFY/pull/ACK round trip entirely.  Default off (0) until measured on
  * hardware — flip with MPIX_DEV_PUSH_MAX=<bytes> (round-2 roadmap item 1). */
+static bool env_flag(const char *name)
+{
+    const char *e = getenv(name);
+    return e && atoi(e);
+}
+
 static uint64_t dev_push_max()
 {
     static const uint64_t v = [] {
@@ -454,21 +460,26 @@ int NativeTransport::init()
             MPIX_ERR("copy stream create failed");
             return -1;
         }
-        /* ONE extra priority stream only: every active hardware queue
-         * beyond ~4 per process oversubscribes the HW scheduler and adds
-         * tens of us to every wait (measured: a 4-plain pool pushed the
-         * 8-B half-RTT from 36.5 to 50 us even for traffic on stream 0) */
-        int lo = 0, hi = 0;
-        if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
-            hi != lo) {
-            if (hipStreamCreateWithPriority(&copy_streams_[1],
-                                            hipStreamNonBlocking, hi) !=
-                hipSuccess) {
+        /* NO standing second stream: every additional hardware queue
+         * (even per another process) oversubscribes the device's HW
+         * scheduler and adds tens of us to every wait — a standing
+         * priority stream alone pushed the 2-process 8-B half-RTT from
+         * 36.5 to ~55 us.  Batched pulls serialize on stream 0 at HBM
+         * rate, so extra streams buy nothing; MPIX_COPY_PRIO_STREAM=1
+         * opts a standing priority stream back in for experiments. */
+        if (env_flag("MPIX_COPY_PRIO_STREAM")) {
+            int lo = 0, hi = 0;
+            if (hipDeviceGetStreamPriorityRange(&lo, &hi) == hipSuccess &&
+                hi != lo) {
+                if (hipStreamCreateWithPriority(&copy_streams_[1],
+                                                hipStreamNonBlocking, hi) !=
+                    hipSuccess) {
+                    (void)hipGetLastError();
+                    copy_streams_[1] = nullptr;
+                }
+            } else {
                 (void)hipGetLastError();
-                copy_streams_[1] = nullptr;
             }
-        } else {
-            (void)hipGetLastError();
         }
         if (hipHostMalloc((void **)&args_h_,
                           (size_t)ARG_SLOTS * MPIX_PULL_BATCH *
