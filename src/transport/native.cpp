@@ -239,15 +239,17 @@ private:
     std::vector<PendingPull> pend_pulls_;
     struct PullBatch {
         hipEvent_t ev;
+        int slot;
         std::vector<PendingPull> items;
     };
     std::list<PullBatch> batches_;
-    /* pinned arg ring: slot i free when no live batch references it */
+    /* pinned arg ring; per-slot busy flags because batches on different
+     * streams (MPIX_COPY_PRIO_STREAM=1) can complete out of order */
     static constexpr int ARG_SLOTS = 8;
     PullArg *args_h_ = nullptr;   /* pinned, ARG_SLOTS * MPIX_PULL_BATCH */
     PullArg *args_d_ = nullptr;
     int arg_slot_ = 0;
-    int arg_slots_live_ = 0;
+    bool arg_busy_[ARG_SLOTS] = {};
 
     /* ---- shm ---- */
     ShmGeom geom_{};
@@ -934,7 +936,7 @@ void NativeTransport::start_dev_copy(InboundMsg &m)
 void NativeTransport::flush_pulls()
 {
     while (!pend_pulls_.empty()) {
-        if (arg_slots_live_ >= ARG_SLOTS) return; /* ring full: next pass */
+        if (arg_busy_[arg_slot_]) return; /* ring full: next pass */
         int n = (int)pend_pulls_.size();
         if (n > MPIX_PULL_BATCH) n = MPIX_PULL_BATCH;
         PullArg *slot_h = args_h_ + (size_t)arg_slot_ * MPIX_PULL_BATCH;
@@ -970,12 +972,13 @@ void NativeTransport::flush_pulls()
         }
         PullBatch b;
         b.ev = get_event();
+        b.slot = arg_slot_;
         (void)hipEventRecord(b.ev, cs);
         b.items.assign(pend_pulls_.begin(), pend_pulls_.begin() + n);
         pend_pulls_.erase(pend_pulls_.begin(), pend_pulls_.begin() + n);
         batches_.push_back(std::move(b));
+        arg_busy_[arg_slot_] = true;
         arg_slot_ = (arg_slot_ + 1) % ARG_SLOTS;
-        arg_slots_live_++;
         if (getenv("MPIX_TRACE") && atoi(getenv("MPIX_TRACE")))
             fprintf(stderr, "[mpix trace] pull batch n=%d total=%lu B\n", n,
                     (unsigned long)total);
@@ -1005,7 +1008,7 @@ int NativeTransport::progress_copies()
             pp.op->ch_status = pp.st;
             pp.op->ch_done.store(1, std::memory_order_release);
         }
-        arg_slots_live_--;
+        arg_busy_[it->slot] = false;
         it = batches_.erase(it);
     }
     for (auto it = copies_.begin(); it != copies_.end();) {

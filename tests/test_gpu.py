@@ -255,6 +255,29 @@ def test_partitioned_ring_device_2proc():
     run_ranks(2, _partitioned_ring_device, timeout=240)
 
 
+def test_many_concurrent_transfers(mpix_env):
+    """150 simultaneous device transfers: exercises multi-batch pull
+    flushes (batch cap 64) and pinned arg-ring wraparound/backpressure."""
+    torch = _torch()
+    mpix = mpix_env
+    n = 150
+    stream = torch.cuda.current_stream()
+    sends = [torch.full((512,), i, dtype=torch.int32, device="cuda")
+             for i in range(n)]
+    recvs = [torch.zeros(512, dtype=torch.int32, device="cuda")
+             for _ in range(n)]
+    reqs = []
+    for i in range(n):
+        reqs.append(mpix.isend_enqueue(sends[i], dest=0, tag=i, stream=stream))
+    for i in range(n):
+        reqs.append(mpix.irecv_enqueue(recvs[i], source=0, tag=i,
+                                       stream=stream))
+    mpix.waitall_enqueue(reqs, stream=stream)
+    torch.cuda.synchronize()
+    for i in range(n):
+        assert (recvs[i] == i).all(), f"transfer {i}"
+
+
 def test_ring_device_4proc_oversubscribed():
     """4 ranks on however many GPUs the box has (1 on CI): exercises
     multi-peer IPC handle exchange and per-pair shm rings — the shape the
