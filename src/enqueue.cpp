@@ -683,14 +683,18 @@ extern "C" int MPIX_Waitall_enqueue(int count, MPIX_Request *reqs,
         return MPI_SUCCESS;
     }
 
-    /* kernel fallback (no memOps, not capturing): one k_wait_and_set per
-     * request.  A single-wave variant polling all flags in one launch
-     * (k_waitall_and_set) exists and passes isolated visibility tests but
-     * hung in situ on ROCm 7.x (store+poll in one divergent wave over the
-     * mapped pool); it is kept and can be forced for experiments with
-     * MPIX_WAITALL_KERNEL=1. */
-    static const bool force_wave =
-        [] { const char *v = getenv("MPIX_WAITALL_KERNEL"); return v && atoi(v); }();
+    /* kernel fallback (no memOps, not capturing).  The single-wave variant
+     * (k_waitall_and_set: one launch polls every flag) is the default for
+     * 4+ requests — its r01 "in-situ hang" was the shared-hardware-queue
+     * blocking class (see add_flag_node), fixed by the copy-stream design;
+     * the forced-mode gpu_ci stage (waitall_wave_kernel) keeps it covered.
+     * MPIX_WAITALL_KERNEL=0 forces per-request k_wait_and_set launches,
+     * =1 forces the wave kernel even for small counts. */
+    static const int wave_env = [] {
+        const char *v = getenv("MPIX_WAITALL_KERNEL");
+        return v && *v ? atoi(v) : -1;
+    }();
+    bool force_wave = wave_env >= 0 ? wave_env != 0 : count >= 4;
     if (!force_wave) {
         for (int i = 0; i < count; i++) {
             Request *req = (Request *)reqs[i];
