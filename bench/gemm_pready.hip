@@ -69,7 +69,202 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define MPIX_GEMM_VARIANT 4
 #endif
 
-#if MPIX_GEMM_VARIANT == 5
+#if MPIX_GEMM_VARIANT == 6
+/* Variant 6 — 8-phase C-quadrant interleave on v4's 256x256 / BK=64 / 2
+ * LDS buffer geometry (the guide's T3+T4 structure: counted vmcnt keeps
+ * the LDS-DMA pipeline alive ACROSS raw barriers; never vmcnt(0) in the
+ * main loop).  One iteration computes K-tiles (2i, 2i+1) in 8 phases;
+ * each phase = one C-quadrant (16 MFMAs) + ONE 16-KiB piece prefetch.
+ *
+ * Piece-death schedule (every wave reads quadrant (qm,qn) at the same
+ * phase, so a region is globally dead one barrier after its last use):
+ *   buf0 (tile 2i):   A-qm0 dies p1, B-qn0 p2, A-qm1 p3, B-qn1 p3
+ *   buf1 (tile 2i+1): A-qm0 dies p5, B-qn0 p6, A-qm1 p7, B-qn1 p7
+ * Prefetch rotation (issue exactly after death of the region it fills):
+ *   p0:A1(2i+1)  p1:B1(2i+1)  p2:A0(2i+2)  p3:B0(2i+2)
+ *   p4:A1(2i+2)  p5:B1(2i+2)  p6:A0(2i+3)  p7:B0(2i+3)
+ * The tightest consumer (p1/p5: B-qn1 issued 4 phases earlier) sets the
+ * uniform wait: s_waitcnt vmcnt(6) (newest 3 pieces = 6 glds stay in
+ * flight) before each barrier; the tail (no more issues) degrades to
+ * vmcnt(0).  Prologue preloads tile0 (4 pieces) + tile1
+ * A0,B0 — exactly what a previous iteration would have issued. */
+#undef BM
+#undef BN
+#undef BK
+#define BM 256
+#define BN 256
+#define BK 64
+#define V4_THREADS 512
+
+__global__ __launch_bounds__(V4_THREADS)
+void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
+                      bf16 *__restrict__ C, int M, int N, int K,
+                      uint32_t *band_cnt, int nparts, int blocks_per_band,
+                      void *dpreq, int publish)
+{
+    __shared__ bf16 lds[2 * 2 * BM * BK];
+    auto As = [&](int b) -> bf16 * { return lds + b * 2 * BM * BK; };
+    auto Bs = [&](int b) -> bf16 * { return As(b) + BM * BK; };
+
+    int nwg = gridDim.x;
+    int wg = blockIdx.x;
+    {
+        int q = nwg / 8, r = nwg % 8, xcd = wg % 8;
+        wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
+    }
+    int tiles_n = N / BN;
+    int tm = wg / tiles_n, tn = wg % tiles_n;
+
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int wm = wave >> 2, wn = wave & 3;
+    int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+    frag4f acc[8][4] = {};
+    const bf16 *Ab = A + (size_t)tm * BM * K;
+    const bf16 *Bb = Bt + (size_t)tn * BN * K;
+
+    /* 16-KiB piece = 16 x 1-KiB LDS chunks (quarters of the operand's
+     * 32-KiB tile image), 2 glds passes x (8 waves x 1 chunk) */
+    static const __device__ uint8_t CH_A0[16] = {0, 1, 2,  3,  4,  5,  6,  7,
+                                                 16, 17, 18, 19, 20, 21, 22, 23};
+    static const __device__ uint8_t CH_A1[16] = {8,  9,  10, 11, 12, 13, 14, 15,
+                                                 24, 25, 26, 27, 28, 29, 30, 31};
+    static const __device__ uint8_t CH_B0[16] = {0, 1, 2,  3,  8,  9,  10, 11,
+                                                 16, 17, 18, 19, 24, 25, 26, 27};
+    static const __device__ uint8_t CH_B1[16] = {4, 5, 6,  7,  12, 13, 14, 15,
+                                                 20, 21, 22, 23, 28, 29, 30, 31};
+
+    auto glds_piece = [&](const bf16 *gbase, bf16 *lbase, int t,
+                          const uint8_t *chunks) {
+        const char *g0 = (const char *)gbase + (size_t)t * BK * sizeof(bf16);
+        #pragma unroll
+        for (int p = 0; p < 2; p++) {
+            unsigned o = ((unsigned)chunks[p * 8 + wave] * 64 +
+                          (unsigned)lane) * 16;
+            unsigned row = o >> 7;
+            unsigned cb = (o & 127u) ^ ((row & 7u) << 4);
+            const void *src = g0 + (size_t)row * K * sizeof(bf16) + cb;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t *)src,
+                (__attribute__((address_space(3))) uint32_t *)
+                    ((char *)lbase + o),
+                16, 0, 0);
+        }
+    };
+    auto frag_at = [&](const bf16 *base, int row, int kb) -> frag8 {
+        int col = kb ^ ((row & 7) << 3);
+        return *(const frag8 *)&base[row * BK + col];
+    };
+    /* one C-quadrant (4x2 fragments) over the tile's full K=64 */
+    auto compute_quad = [&](int buf, int qm, int qn) {
+        #pragma unroll
+        for (int kh = 0; kh < 2; kh++) {
+            int kb = kh * 32 + lk8;
+            frag8 af[4], bf[2];
+            #pragma unroll
+            for (int i = 0; i < 4; i++)
+                af[i] = frag_at(As(buf),
+                                wm * 128 + qm * 64 + i * 16 + lrow, kb);
+            #pragma unroll
+            for (int j = 0; j < 2; j++)
+                bf[j] = frag_at(Bs(buf),
+                                wn * 64 + qn * 32 + j * 16 + lrow, kb);
+            #pragma unroll
+            for (int i = 0; i < 4; i++)
+                #pragma unroll
+                for (int j = 0; j < 2; j++)
+                    acc[qm * 4 + i][qn * 2 + j] =
+                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af[i], bf[j], acc[qm * 4 + i][qn * 2 + j], 0, 0,
+                            0);
+        }
+    };
+
+    int tiles = K / BK; /* host asserts tiles even and >= 4 for this variant */
+    /* prologue = what a previous iteration would have left in flight */
+    glds_piece(Ab, As(0), 0, CH_A0);
+    glds_piece(Bb, Bs(0), 0, CH_B0);
+    glds_piece(Ab, As(0), 0, CH_A1);
+    glds_piece(Bb, Bs(0), 0, CH_B1);
+    glds_piece(Ab, As(1), 1, CH_A0);
+    glds_piece(Bb, Bs(1), 1, CH_B0);
+
+    for (int t2 = 0; t2 < tiles; t2 += 2) {
+        bool tail = (t2 + 3 >= tiles);
+        #pragma unroll
+        for (int p = 0; p < 8; p++) {
+            /* leave newest 3 pieces (6 glds) in flight: the tightest
+             * consumer (p1/p5: B-qn1 issued 4 phases earlier) still
+             * retires under vmcnt(6); everything else is >=5 deep */
+            if (!tail)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            /* prefetch rotation (see header comment) */
+            switch (p) {
+            case 0:
+                glds_piece(Ab, As(1), t2 + 1, CH_A1);
+                break;
+            case 1:
+                glds_piece(Bb, Bs(1), t2 + 1, CH_B1);
+                break;
+            case 2:
+                if (t2 + 2 < tiles) glds_piece(Ab, As(0), t2 + 2, CH_A0);
+                break;
+            case 3:
+                if (t2 + 2 < tiles) glds_piece(Bb, Bs(0), t2 + 2, CH_B0);
+                break;
+            case 4:
+                if (t2 + 2 < tiles) glds_piece(Ab, As(0), t2 + 2, CH_A1);
+                break;
+            case 5:
+                if (t2 + 2 < tiles) glds_piece(Bb, Bs(0), t2 + 2, CH_B1);
+                break;
+            case 6:
+                if (t2 + 3 < tiles) glds_piece(Ab, As(1), t2 + 3, CH_A0);
+                break;
+            case 7:
+                if (t2 + 3 < tiles) glds_piece(Bb, Bs(1), t2 + 3, CH_B0);
+                break;
+            }
+            compute_quad(p >> 2, (p >> 1) & 1, p & 1);
+        }
+    }
+
+    size_t crow0 = (size_t)tm * BM + wm * 128;
+    size_t ccol0 = (size_t)tn * BN + wn * 64;
+    #pragma unroll
+    for (int i = 0; i < 8; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            size_t col = ccol0 + j * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; r++) {
+                size_t row = crow0 + i * 16 + (lane >> 4) * 4 + r;
+                C[row * N + col] = (bf16)acc[i][j][r];
+            }
+        }
+
+    if (!publish) return;
+    int band = (int)((size_t)tm * BM * (size_t)nparts / M);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t prev = __hip_atomic_fetch_add(&band_cnt[band], 1,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == (uint32_t)blocks_per_band - 1) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+            MPIX_Pready(band, dpreq);
+        }
+    }
+}
+
+#elif MPIX_GEMM_VARIANT == 5
 /* Variant 5 — v4's 256x256 glds structure pushed to the guide's pipelined
  * tier: BK=32 tiles in FOUR LDS buffers (4 x 32 KiB), two tiles in flight
  * across RAW barriers with counted vmcnt (never vmcnt(0) in the loop —
@@ -701,6 +896,9 @@ int main(int argc, char **argv)
     CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0);
     /* one band >= one tile row; clamp so small M still works */
     int nparts = check ? M / BM : (M / BM < NPARTS ? M / BM : NPARTS);
+#if MPIX_GEMM_VARIANT == 6
+    CHECK(K % (2 * BK) == 0 && K >= 4 * BK); /* 8-phase pair schedule */
+#endif
     CHECK(M % nparts == 0 && (M / nparts) % BM == 0);
     int blocks_per_band = (M / nparts / BM) * (N / BN);
 
