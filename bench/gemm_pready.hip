@@ -404,6 +404,8 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
 }
 
 #elif MPIX_GEMM_VARIANT == 4
+/* (optional -DMPIX_GEMM_GROUP=G: grouped block order — consecutive
+ * workgroups cover a tiles_m x G super-column so A rows re-hit L2) */
 /* Variant 4 — guide §5 "glds + 2 LDS buffers + BK=64" structure on a
  * 256x256 tile (the measured ~1.1-1.2 PF tier for this shape):
  *  - 512 threads = 8 waves in a 2(M)x4(N) grid, each owning a 128x64
@@ -444,7 +446,20 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
         wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + wg / 8;
     }
     int tiles_n = N / BN;
-    int tm = wg / tiles_n, tn = wg % tiles_n;
+    int tm, tn;
+#ifdef MPIX_GEMM_GROUP
+    {
+        int tiles_m = M / BM;
+        int per_group = tiles_m * MPIX_GEMM_GROUP;
+        int group = wg / per_group, rem = wg % per_group;
+        tn = group * MPIX_GEMM_GROUP + rem % MPIX_GEMM_GROUP;
+        tm = rem / MPIX_GEMM_GROUP;
+        if (tn >= tiles_n) { tn = wg % tiles_n; tm = wg / tiles_n; }
+    }
+#else
+    tm = wg / tiles_n;
+    tn = wg % tiles_n;
+#endif
 
     int tid = threadIdx.x;
     int wave = tid >> 6, lane = tid & 63;
