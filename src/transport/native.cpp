@@ -952,9 +952,13 @@ void NativeTransport::flush_pulls()
         }
         hipStream_t cs = copy_stream(total);
         unsigned threads = 256;
+        static const unsigned max_blocks = [] {
+            const char *e = getenv("MPIX_PULL_BLOCKS");
+            return e && atoi(e) > 0 ? (unsigned)atoi(e) : 512u;
+        }();
         unsigned blocks = (unsigned)((total / 16 + threads - 1) / threads);
         if (blocks == 0) blocks = 1;
-        if (blocks > 512) blocks = 512;
+        if (blocks > max_blocks) blocks = max_blocks;
         hipLaunchKernelGGL(k_pull_multi, dim3(blocks), dim3(threads), 0, cs,
                            slot_d, n);
         hipError_t e = hipGetLastError();
