@@ -954,7 +954,9 @@ void NativeTransport::flush_pulls()
         unsigned threads = 256;
         static const unsigned max_blocks = [] {
             const char *e = getenv("MPIX_PULL_BLOCKS");
-            return e && atoi(e) > 0 ? (unsigned)atoi(e) : 512u;
+            /* A/B @256 MiB batches: 512 -> 1411 GB/s, 1024 -> 1700,
+             * 2048 -> 1594 (4 blocks/CU fills HBM without thrash) */
+            return e && atoi(e) > 0 ? (unsigned)atoi(e) : 1024u;
         }();
         unsigned blocks = (unsigned)((total / 16 + threads - 1) / threads);
         if (blocks == 0) blocks = 1;
