@@ -46,8 +46,8 @@ def test_c_suite_2rank(binary):
 
 
 @pytest.mark.parametrize("binary", ["ring", "ring_all", "ring_partitioned",
-                                    "ring_subcomm", "ring_partitioned_subcomm",
-                                    "ring_all_device"])
+                                    "ring_subcomm",
+                                    "ring_partitioned_subcomm"])
 def test_c_suite_4rank(binary):
     _ensure_built()
     assert "PASS" in _run(binary, 4)
