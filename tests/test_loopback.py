@@ -174,3 +174,36 @@ def test_partitioned_out_of_order_pready(mpix_env):
     assert (recv == send).all()
     mpix.request_free(ps)
     mpix.request_free(pr)
+
+
+def test_wildcard_recv_ignores_partitioned(mpix_env):
+    """A wildcard basic receive must never match partition messages —
+    partitioned traffic matches in its own channel (src/transport/
+    native.cpp match(): want_part != is_part excludes cross-matching)."""
+    import numpy as np
+    mpix = mpix_env
+    parts, per = 4, 64
+    psend = np.arange(parts * per, dtype=np.int32)
+    precv = np.zeros_like(psend)
+    wild = np.full(per, -1, dtype=np.int32)
+    # wildcard basic recv posted FIRST, same tag as the partitioned pair
+    rw = mpix.irecv_enqueue(wild, source=mpix.ANY_SOURCE, tag=mpix.ANY_TAG)
+    ps = mpix.psend_init(psend, parts, dest=0, tag=5)
+    pr = mpix.precv_init(precv, parts, source=0, tag=5)
+    mpix.start(pr)
+    mpix.start(ps)
+    for p in range(parts):
+        mpix.pready(p, ps)
+    mpix.wait(pr)
+    mpix.wait(ps)
+    assert (precv == psend).all()
+    assert (wild == -1).all(), "wildcard recv stole a partition message"
+    # the wildcard recv still works for a real basic message
+    basic = np.full(per, 77, dtype=np.int32)
+    rs = mpix.isend_enqueue(basic, dest=0, tag=99)
+    st = mpix.wait(rw)
+    mpix.wait(rs)
+    assert (wild == 77).all()
+    assert st["tag"] == 99
+    mpix.request_free(ps)
+    mpix.request_free(pr)
