@@ -77,8 +77,10 @@ __global__ void k_pull_multi(const PullArg *__restrict__ args, int ncopies)
 
 /* Sender-push threshold for small DEVICE payloads: stage D2H into the shm
  * chunk ring like a host message (receiver H2D's it out), skipping the
- * NOTIFY/pull/ACK round trip entirely.  Default off (0) until measured on
- * hardware — flip with MPIX_DEV_PUSH_MAX=<bytes> (round-2 roadmap item 1). */
+ * NOTIFY/pull/ACK round trip.  Measured WORSE (74-97 us vs ~38 us half-RTT:
+ * the two proxy-synchronous staging copies cost more than the round trip
+ * they remove — profiles/r02_pingpong_devpush.json); kept off, tunable for
+ * experiments with MPIX_DEV_PUSH_MAX=<bytes>. */
 static bool env_flag(const char *name)
 {
     const char *e = getenv(name);
