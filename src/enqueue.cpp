@@ -79,9 +79,11 @@ __global__ void k_waitall_and_set(uint32_t *flags, const int32_t *idx,
 /* ------------------------------------------------------------------ helpers */
 
 /* Record that a spin-wait kernel has been emitted somewhere (stream
- * fallback, capture, or graph node) — flips the transport to SDMA copies
- * so no copy kernel can ever be queued behind a spinning wave (see
- * pull_kernels_safe in transport/native.cpp). */
+ * fallback, capture, or graph node): the transport migrates its copy
+ * stream to a greatest-priority one, whose hardware queue no user stream
+ * (and hence no spinning wave) can share — see copy_stream() in
+ * transport/native.cpp.  The store is ordered before any launch of the
+ * kernel it reports. */
 static inline void mark_spin_wait()
 {
     g_state->spin_wait_kernels.store(true, std::memory_order_release);
