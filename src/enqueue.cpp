@@ -511,8 +511,8 @@ extern "C" int MPIX_Wait_enqueue(MPIX_Request *reqp, MPI_Status *status,
                                              (uint32_t)MPIX_FLAG_CLEANUP, 0));
     } else {
         mark_spin_wait();
-            hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0, stream, flag_d,
-                           (uint32_t)MPIX_FLAG_COMPLETED,
+        hipLaunchKernelGGL(k_wait_and_set, dim3(1), dim3(1), 0, stream,
+                           flag_d, (uint32_t)MPIX_FLAG_COMPLETED,
                            (uint32_t)MPIX_FLAG_CLEANUP);
         MPIX_CHECK_HIP(hipGetLastError());
     }
