@@ -152,9 +152,9 @@ static int attach_cleanup(hipGraph_t graph, Request *req)
  * same hardware queue as the transport's copy work (graph execution
  * serializes queues with AQL barrier packets) — observed deterministically
  * after a few stream-create cycles (gpurun_out/diag2_*).  The kernel-node
- * fallback remains for memOps-less configs; emitting one flips the
- * transport to SDMA-only copies (mark_spin_wait) so the blocked-copy cycle
- * cannot form. */
+ * fallback remains for memOps-less configs; emitting one migrates the
+ * transport's copy stream to its own priority queue (mark_spin_wait) so
+ * the blocked-copy cycle cannot form. */
 static int add_flag_node(hipGraph_t g, bool is_wait, uint32_t *flag_d,
                          uint32_t val)
 {
