@@ -454,12 +454,25 @@ void gemm_bf16_pready(const bf16 *__restrict__ A, const bf16 *__restrict__ Bt,
     int tm, tn;
 #if MPIX_GEMM_GROUP > 0
     {
+        /* bijective grouped order: full G-wide column groups first, then a
+         * (tiles_n % G)-wide tail group — a naive modulo fallback silently
+         * COLLIDED for tiles_n % G != 0 (two wgs computing one C tile,
+         * another tile never written; mirrored in the host bijectivity
+         * test tests/test_units.py::test_gemm_group_mapping_bijective) */
         int tiles_m = M / BM;
-        int per_group = tiles_m * MPIX_GEMM_GROUP;
-        int group = wg / per_group, rem = wg % per_group;
-        tn = group * MPIX_GEMM_GROUP + rem % MPIX_GEMM_GROUP;
-        tm = rem / MPIX_GEMM_GROUP;
-        if (tn >= tiles_n) { tn = wg % tiles_n; tm = wg / tiles_n; }
+        int gcols = (tiles_n / MPIX_GEMM_GROUP) * MPIX_GEMM_GROUP;
+        int ngrouped = tiles_m * gcols;
+        if (wg < ngrouped) {
+            int per_group = tiles_m * MPIX_GEMM_GROUP;
+            int group = wg / per_group, rem = wg % per_group;
+            tn = group * MPIX_GEMM_GROUP + rem % MPIX_GEMM_GROUP;
+            tm = rem / MPIX_GEMM_GROUP;
+        } else {
+            int tail = tiles_n - gcols; /* >= 1 when any wg lands here */
+            int r = wg - ngrouped;
+            tn = gcols + r % tail;
+            tm = r / tail;
+        }
     }
 #else
     tm = wg / tiles_n;

@@ -258,3 +258,38 @@ def test_trace_and_stats_emit(monkeypatch):
     assert "[mpix trace" in r.stderr
     assert "[mpix stats" in r.stderr
     assert "latency histogram" in r.stderr
+
+
+def test_gemm_group_mapping_bijective():
+    """Host replica of bench/gemm_pready.hip's grouped + XCD block-order
+    mapping (variant 4): the composition must be a bijection over the
+    grid for EVERY (tiles_m, tiles_n) — the original overflow fallback
+    silently collided when tiles_n % G != 0 (a C tile computed twice,
+    another never written)."""
+    def xcd(wg, nwg):
+        q, r = nwg // 8, nwg % 8
+        x = wg % 8
+        return (x * (q + 1) if x < r else r * (q + 1) + (x - r) * q) + wg // 8
+
+    def grouped(wg, tiles_m, tiles_n, G):
+        gcols = (tiles_n // G) * G
+        ngrouped = tiles_m * gcols
+        if wg < ngrouped:
+            per_group = tiles_m * G
+            group, rem = divmod(wg, per_group)
+            return rem // G, group * G + rem % G
+        tail = tiles_n - gcols
+        r = wg - ngrouped
+        return r // tail, gcols + r % tail
+
+    G = 4
+    for tiles_m in range(1, 12):
+        for tiles_n in range(1, 12):
+            nwg = tiles_m * tiles_n
+            seen = set()
+            for wg0 in range(nwg):
+                tm, tn = grouped(xcd(wg0, nwg), tiles_m, tiles_n, G)
+                assert 0 <= tm < tiles_m and 0 <= tn < tiles_n, \
+                    (tiles_m, tiles_n, wg0, tm, tn)
+                seen.add((tm, tn))
+            assert len(seen) == nwg, (tiles_m, tiles_n, len(seen))
